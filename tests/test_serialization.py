@@ -1,0 +1,97 @@
+import pytest
+import torch
+
+from torchsnapshot_amd.serialization import (
+    SERIALIZER_BUFFER,
+    SERIALIZER_TORCH_SAVE,
+    dtype_to_str,
+    pick_serializer,
+    str_to_dtype,
+    tensor_as_memoryview,
+    tensor_from_memoryview,
+    torch_load_from_bytes,
+    torch_save_as_bytes,
+)
+from torchsnapshot_amd.test_utils import rand_tensor, tensor_eq
+
+_PLAIN_DTYPES = [
+    torch.float32,
+    torch.float64,
+    torch.float16,
+    torch.bfloat16,
+    torch.complex64,
+    torch.complex128,
+    torch.uint8,
+    torch.int8,
+    torch.int16,
+    torch.int32,
+    torch.int64,
+    torch.bool,
+    torch.float8_e4m3fn,
+    torch.float8_e5m2,
+]
+
+
+@pytest.mark.parametrize("dtype", _PLAIN_DTYPES, ids=str)
+def test_buffer_round_trip(dtype):
+    t = rand_tensor((16, 9), dtype=dtype)
+    mv = tensor_as_memoryview(t)
+    assert mv.nbytes == t.numel() * t.element_size()
+    # writable copy simulates a read buffer
+    buf = bytearray(mv)
+    t2 = tensor_from_memoryview(memoryview(buf), dtype=dtype, shape=(16, 9))
+    assert tensor_eq(t, t2)
+
+
+def test_dtype_string_round_trip():
+    for dtype in _PLAIN_DTYPES + [torch.qint8, torch.quint8, torch.qint32]:
+        assert str_to_dtype(dtype_to_str(dtype)) == dtype
+
+
+def test_pick_serializer():
+    assert pick_serializer(torch.rand(3)) == SERIALIZER_BUFFER
+    assert pick_serializer(rand_tensor((3,), torch.qint8)) == SERIALIZER_TORCH_SAVE
+
+
+def test_scalar_tensor():
+    t = torch.tensor(4.25, dtype=torch.bfloat16)
+    mv = tensor_as_memoryview(t)
+    t2 = tensor_from_memoryview(memoryview(bytearray(mv)), torch.bfloat16, ())
+    assert t2.item() == 4.25
+
+
+def test_empty_tensor():
+    t = torch.empty(0, 5)
+    mv = tensor_as_memoryview(t)
+    assert mv.nbytes == 0
+    t2 = tensor_from_memoryview(memoryview(b""), torch.float32, (0, 5))
+    assert t2.shape == (0, 5)
+
+
+def test_non_contiguous_rejected():
+    t = torch.rand(4, 4).t()
+    with pytest.raises(ValueError):
+        tensor_as_memoryview(t)
+
+
+def test_torch_save_round_trip():
+    obj = {"a": torch.rand(3), "b": [1, 2]}
+    data = torch_save_as_bytes(obj)
+    obj2 = torch_load_from_bytes(data)
+    assert torch.equal(obj["a"], obj2["a"])
+    assert obj2["b"] == [1, 2]
+
+
+def test_quantized_torch_save_round_trip():
+    t = rand_tensor((8, 8), torch.qint8)
+    t2 = torch_load_from_bytes(torch_save_as_bytes(t))
+    assert tensor_eq(t, t2)
+
+
+def test_memoryview_zero_copy():
+    t = torch.rand(128)
+    mv = tensor_as_memoryview(t)
+    t[0] = 42.0
+    # zero-copy: mutation visible through the view
+    t2 = tensor_from_memoryview(mv, torch.float32, (128,))
+    assert t2[0].item() == 42.0

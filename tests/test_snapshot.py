@@ -1,0 +1,203 @@
+"""End-to-end single-process take/restore tests (CPU)."""
+
+import os
+
+import pytest
+import torch
+
+from torchsnapshot_amd import RNGState, Snapshot, StateDict
+from torchsnapshot_amd.test_utils import (
+    check_state_dict_eq,
+    rand_tensor,
+    tmp_snapshot_path,
+)
+
+
+class _Model(torch.nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.lin1 = torch.nn.Linear(17, 33)
+        self.lin2 = torch.nn.Linear(33, 5)
+        self.register_buffer("buf", torch.rand(8))
+
+    def forward(self, x):
+        return self.lin2(torch.relu(self.lin1(x)))
+
+
+def test_take_restore_module(toggle_batching):
+    model = _Model()
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"model": model})
+        assert os.path.exists(os.path.join(path, ".snapshot_metadata"))
+        model2 = _Model()
+        assert not check_state_dict_eq(model.state_dict(), model2.state_dict())
+        snapshot.restore({"model": model2})
+        assert check_state_dict_eq(model.state_dict(), model2.state_dict())
+
+
+def test_take_restore_module_and_optimizer():
+    model = _Model()
+    optim = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    # generate optimizer state
+    model(torch.rand(4, 17)).sum().backward()
+    optim.step()
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"model": model, "optim": optim})
+        model2 = _Model()
+        optim2 = torch.optim.SGD(model2.parameters(), lr=0.1, momentum=0.9)
+        model2(torch.rand(4, 17)).sum().backward()
+        optim2.step()
+        snapshot.restore({"model": model2, "optim": optim2})
+        assert check_state_dict_eq(model.state_dict(), model2.state_dict())
+        assert check_state_dict_eq(optim.state_dict(), optim2.state_dict())
+
+
+def test_primitives_and_objects():
+    progress = StateDict(
+        epoch=7,
+        step=123,
+        lr=0.125,
+        name="run-42",
+        done=False,
+        blob=b"\x00\x01",
+        nested={"a": [1, 2, 3], "t": torch.arange(5)},
+        custom=(1, 2),  # tuple -> object fallback
+    )
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"progress": progress})
+        progress2 = StateDict()
+        snapshot.restore({"progress": progress2})
+        assert progress2["epoch"] == 7
+        assert progress2["step"] == 123
+        assert progress2["lr"] == 0.125
+        assert progress2["name"] == "run-42"
+        assert progress2["done"] is False
+        assert progress2["blob"] == b"\x00\x01"
+        assert torch.equal(progress2["nested"]["t"], torch.arange(5))
+        assert progress2["nested"]["a"] == [1, 2, 3]
+        assert progress2["custom"] == (1, 2)
+
+
+def test_all_dtypes(toggle_batching):
+    dtypes = [
+        torch.float32,
+        torch.float64,
+        torch.float16,
+        torch.bfloat16,
+        torch.complex64,
+        torch.uint8,
+        torch.int8,
+        torch.int16,
+        torch.int32,
+        torch.int64,
+        torch.bool,
+        torch.qint8,
+        torch.quint8,
+    ]
+    sd = StateDict({str(d): rand_tensor((7, 3), d) for d in dtypes})
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        sd2 = StateDict()
+        snapshot.restore({"sd": sd2})
+        assert check_state_dict_eq(sd.state_dict(), sd2.state_dict())
+
+
+def test_chunked_tensor(toggle_chunking):
+    big = torch.rand(1000, 10)
+    sd = StateDict(big=big)
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        sd2 = StateDict(big=torch.zeros(1000, 10))
+        snapshot.restore({"sd": sd2})
+        assert torch.equal(sd2["big"], big)
+
+
+def test_rng_state_invariance():
+    torch.manual_seed(777)
+    rng = RNGState()
+    with tmp_snapshot_path() as path:
+        state_before = torch.get_rng_state()
+        snapshot = Snapshot.take(path, {"rng": rng, "x": StateDict(v=1)})
+        # taking the snapshot must not perturb the RNG stream
+        assert torch.equal(torch.get_rng_state(), state_before)
+        draw_after_take = torch.rand(3)
+        # restore, then the same draw must reproduce
+        snapshot.restore({"rng": rng})
+        assert torch.equal(torch.rand(3), draw_after_take)
+
+
+def test_restore_into_different_dtype():
+    sd = StateDict(t=torch.rand(5, dtype=torch.float32))
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        sd2 = StateDict(t=torch.zeros(5, dtype=torch.float64))
+        snapshot.restore({"sd": sd2})
+        assert torch.allclose(sd2["t"].float(), sd["t"])
+        assert sd2["t"].dtype == torch.float64
+
+
+def test_metadata_missing_raises():
+    with tmp_snapshot_path() as path:
+        os.makedirs(path, exist_ok=True)
+        with pytest.raises(RuntimeError, match="incomplete or corrupted"):
+            _ = Snapshot(path).metadata
+
+
+def test_get_manifest():
+    sd = StateDict(a=1, t=torch.rand(3))
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        manifest = snapshot.get_manifest()
+        assert "0/sd/a" in manifest
+        assert manifest["0/sd/t"]["kind"] == "tensor"
+
+
+def test_non_contiguous_and_view_tensors():
+    base = torch.rand(10, 10)
+    sd = StateDict(
+        transposed=base.t(),
+        narrow=base[2:5],
+        strided=base[::2, ::3],
+        expanded=torch.rand(1, 4).expand(3, 4),
+    )
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        sd2 = StateDict(
+            transposed=torch.zeros(10, 10),
+            narrow=torch.zeros(3, 10),
+            strided=torch.zeros(5, 4),
+            expanded=torch.zeros(3, 4),
+        )
+        snapshot.restore({"sd": sd2})
+        assert torch.equal(sd2["transposed"], base.t())
+        assert torch.equal(sd2["narrow"], base[2:5])
+        assert torch.equal(sd2["strided"], base[::2, ::3])
+        assert torch.equal(sd2["expanded"], torch.rand(1, 4).expand(3, 4) * 0 + sd["expanded"])
+
+
+def test_read_object():
+    sd = StateDict(t=torch.rand(17), n=42)
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        t = snapshot.read_object("0/sd/t")
+        assert torch.equal(t, sd["t"])
+        assert snapshot.read_object("0/sd/n") == 42
+        out = torch.zeros(17)
+        snapshot.read_object("0/sd/t", obj_out=out)
+        assert torch.equal(out, sd["t"])
+
+
+def test_read_object_tiled():
+    sd = StateDict(t=torch.rand(1000, 100))
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        t = snapshot.read_object("0/sd/t", memory_budget_bytes=64 * 1024)
+        assert torch.equal(t, sd["t"])
+
+
+def test_get_state_dict_for_key():
+    model = _Model()
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"model": model})
+        sd = snapshot.get_state_dict_for_key("model")
+        assert check_state_dict_eq(dict(model.state_dict()), dict(sd))

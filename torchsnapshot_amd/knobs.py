@@ -1,0 +1,141 @@
+"""Tunable knobs, overridable via environment variables.
+
+Parity with reference torchsnapshot/knobs.py:23-132 (env-var constants +
+context-manager overrides for tests), re-tuned for MI355X nodes:
+
+- The per-GPU HBM3E pool is 288 GB and host DDR on an MI355X node is large,
+  so default chunk/shard/slab sizes are bigger than the reference's
+  (512 MB/512 MB/128 MB): larger slabs mean fewer, larger NVMe writes and
+  fewer D2H launches over the PCIe Gen5 link (~55-60 GB/s effective).
+"""
+
+from __future__ import annotations
+
+import contextlib
+import os
+from typing import Iterator
+
+_MB = 1024 * 1024
+
+
+def _env_bytes(name: str, default: int) -> int:
+    val = os.environ.get(name)
+    if val is None:
+        return default
+    return int(float(val))
+
+
+def _env_int(name: str, default: int) -> int:
+    val = os.environ.get(name)
+    if val is None:
+        return default
+    return int(val)
+
+
+def _env_flag(name: str) -> bool:
+    return os.environ.get(name, "0") not in ("0", "", "false", "False")
+
+
+# -- write-path sizing -------------------------------------------------------
+
+def get_max_chunk_size_bytes() -> int:
+    """Tensors larger than this are split into chunks for pipelined I/O."""
+    return _env_bytes("TSAMD_MAX_CHUNK_SIZE_BYTES", 512 * _MB)
+
+
+def get_max_shard_size_bytes() -> int:
+    """Local shards of sharded tensors are subdivided to at most this size."""
+    return _env_bytes("TSAMD_MAX_SHARD_SIZE_BYTES", 512 * _MB)
+
+
+def get_slab_size_threshold_bytes() -> int:
+    """Write requests smaller than this are packed into batched slabs."""
+    return _env_bytes("TSAMD_SLAB_SIZE_THRESHOLD_BYTES", 256 * _MB)
+
+
+def is_batching_disabled() -> bool:
+    return _env_flag("TSAMD_DISABLE_BATCHING")
+
+
+def is_partitioner_disabled() -> bool:
+    return _env_flag("TSAMD_DISABLE_PARTITIONER")
+
+
+# -- execution ---------------------------------------------------------------
+
+def get_max_io_concurrency() -> int:
+    """Maximum concurrent storage I/O operations per rank."""
+    return _env_int("TSAMD_MAX_PER_RANK_IO_CONCURRENCY", 16)
+
+
+def get_num_staging_threads() -> int:
+    """Executor threads used for CPU-side staging (GIL-releasing copies)."""
+    return _env_int("TSAMD_NUM_STAGING_THREADS", 4)
+
+
+def get_memory_budget_override_bytes() -> int | None:
+    val = os.environ.get("TSAMD_PER_RANK_MEMORY_BUDGET_BYTES")
+    return None if val is None else int(float(val))
+
+
+def get_pinned_block_size_bytes() -> int:
+    """Size of one pinned host staging block in the D2H ring."""
+    return _env_bytes("TSAMD_PINNED_BLOCK_SIZE_BYTES", 256 * _MB)
+
+
+def get_pinned_block_count() -> int:
+    """Number of pinned host staging blocks in the D2H ring."""
+    return _env_int("TSAMD_PINNED_BLOCK_COUNT", 4)
+
+
+def is_hip_staging_disabled() -> bool:
+    """Force the torch fallback for device staging (debug only)."""
+    return _env_flag("TSAMD_DISABLE_HIP_STAGING")
+
+
+def get_storage_write_chunk_bytes() -> int:
+    """Chunk size for filesystem pwrite calls (large sequential writes)."""
+    return _env_bytes("TSAMD_FS_WRITE_CHUNK_BYTES", 64 * _MB)
+
+
+# -- elasticity --------------------------------------------------------------
+
+def is_sharded_elasticity_root_only() -> bool:
+    """If set, sharded-tensor elasticity handling only applies at state-dict
+    root level (mirrors the reference's narrow-scope knob)."""
+    return _env_flag("TSAMD_ENABLE_SHARDED_TENSOR_ELASTICITY_ROOT_ONLY")
+
+
+# -- test overrides ----------------------------------------------------------
+
+@contextlib.contextmanager
+def override_env(name: str, value: str) -> Iterator[None]:
+    prev = os.environ.get(name)
+    os.environ[name] = value
+    try:
+        yield
+    finally:
+        if prev is None:
+            os.environ.pop(name, None)
+        else:
+            os.environ[name] = prev
+
+
+def override_max_chunk_size_bytes(nbytes: int):
+    return override_env("TSAMD_MAX_CHUNK_SIZE_BYTES", str(nbytes))
+
+
+def override_max_shard_size_bytes(nbytes: int):
+    return override_env("TSAMD_MAX_SHARD_SIZE_BYTES", str(nbytes))
+
+
+def override_slab_size_threshold_bytes(nbytes: int):
+    return override_env("TSAMD_SLAB_SIZE_THRESHOLD_BYTES", str(nbytes))
+
+
+def override_batching_disabled(disabled: bool):
+    return override_env("TSAMD_DISABLE_BATCHING", "1" if disabled else "0")
+
+
+def override_max_io_concurrency(n: int):
+    return override_env("TSAMD_MAX_PER_RANK_IO_CONCURRENCY", str(n))

@@ -1,0 +1,372 @@
+"""Pipelined execution of write/read requests under a host-memory budget.
+
+Design (differs from the reference's nested-event-loop scheduler,
+torchsnapshot/scheduler.py:222-446): every pipeline runs on a dedicated
+background thread with its own asyncio event loop. One coroutine per
+request drives stage -> write (or read -> consume); a budget condition
+variable bounds the total bytes of in-flight buffers, and a semaphore
+bounds storage concurrency. Because the caller thread never runs the loop,
+the same machinery serves sync take (wait for everything), async_take
+(wait for staging only — the thread keeps draining storage I/O), and
+nested/Jupyter callers without re-entrant-loop tricks.
+
+Collectives never run on the pipeline thread (RCCL/process-group calls stay
+on the caller thread), matching the constraint the reference documents at
+snapshot.py:1010.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import socket
+import threading
+import time
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass, field
+from typing import Awaitable, Callable, List, Optional, TypeVar
+
+import psutil
+
+from . import knobs
+from .io_types import ReadIO, ReadReq, StageContext, StoragePlugin, WriteIO, WriteReq
+from .pg_wrapper import PGWrapper
+
+logger = logging.getLogger(__name__)
+
+T = TypeVar("T")
+
+_MAX_PER_RANK_MEMORY_BUDGET_BYTES = 32 * 1024 * 1024 * 1024
+_AVAILABLE_MEMORY_FRACTION = 0.6
+
+
+def run_coro_sync(coro: Awaitable[T]) -> T:
+    """Run a coroutine to completion from sync code, safely even when the
+    caller is already inside a running event loop (e.g. Jupyter)."""
+    try:
+        asyncio.get_running_loop()
+    except RuntimeError:
+        return asyncio.run(coro)  # type: ignore[arg-type]
+    with ThreadPoolExecutor(max_workers=1) as ex:
+        return ex.submit(asyncio.run, coro).result()  # type: ignore[arg-type]
+
+
+def get_local_world_size(pg: PGWrapper) -> int:
+    """Number of ranks on this host (divides the host-memory budget)."""
+    hostnames: List[Optional[str]] = [None] * pg.get_world_size()
+    pg.all_gather_object(hostnames, socket.gethostname())
+    return hostnames.count(socket.gethostname())
+
+
+def get_process_memory_budget_bytes(pg: PGWrapper) -> int:
+    override = knobs.get_memory_budget_override_bytes()
+    if override is not None:
+        logger.info("Manual memory budget: %d bytes", override)
+        return override
+    available = psutil.virtual_memory().available
+    budget = int(
+        available * _AVAILABLE_MEMORY_FRACTION / max(get_local_world_size(pg), 1)
+    )
+    return min(budget, _MAX_PER_RANK_MEMORY_BUDGET_BYTES)
+
+
+@dataclass
+class ExecutionStats:
+    total_reqs: int = 0
+    staged_reqs: int = 0
+    done_reqs: int = 0
+    staged_bytes: int = 0
+    io_bytes: int = 0
+    begin_ts: float = field(default_factory=time.monotonic)
+    staged_ts: Optional[float] = None
+    end_ts: Optional[float] = None
+
+    def throughput_bytes_per_sec(self) -> float:
+        end = self.end_ts or time.monotonic()
+        dur = max(end - self.begin_ts, 1e-9)
+        return self.io_bytes / dur
+
+
+class _Budget:
+    """Async counter of in-flight buffer bytes with an escape hatch: a
+    request costing more than the whole budget may run when the pipeline is
+    otherwise empty (so oversized items make progress instead of
+    deadlocking)."""
+
+    def __init__(self, limit: int) -> None:
+        self.limit = limit
+        self.in_use = 0
+        self.cond = asyncio.Condition()
+
+    async def acquire(self, cost: int) -> None:
+        async with self.cond:
+            while not (
+                self.in_use + cost <= self.limit
+                or (self.in_use == 0 and cost > self.limit)
+            ):
+                await self.cond.wait()
+            self.in_use += cost
+
+    async def release(self, cost: int) -> None:
+        async with self.cond:
+            self.in_use -= cost
+            self.cond.notify_all()
+
+
+class PendingIOWork:
+    """Handle to an in-flight pipeline running on its own thread."""
+
+    def __init__(
+        self,
+        thread: threading.Thread,
+        staged_event: threading.Event,
+        done_event: threading.Event,
+        stats: ExecutionStats,
+        exc_holder: List[BaseException],
+    ) -> None:
+        self._thread = thread
+        self._staged_event = staged_event
+        self._done_event = done_event
+        self.stats = stats
+        self._exc_holder = exc_holder
+
+    def _maybe_raise(self) -> None:
+        if self._exc_holder:
+            raise self._exc_holder[0]
+
+    def wait_staged(self) -> None:
+        """Block until all buffers are staged in host memory (storage I/O may
+        still be in flight). After this, the source tensors are safe to
+        mutate."""
+        self._staged_event.wait()
+        if not self._done_event.is_set():
+            # staging done; errors so far would have set done as well
+            if self._exc_holder:
+                self._thread.join()
+                self._maybe_raise()
+        else:
+            self._maybe_raise()
+
+    def complete(self) -> None:
+        """Block until all storage I/O finished; re-raise pipeline errors."""
+        self._thread.join()
+        self._maybe_raise()
+
+    def done(self) -> bool:
+        return self._done_event.is_set()
+
+
+def _spawn_pipeline(
+    main: Callable[[], Awaitable[None]],
+    stats: ExecutionStats,
+    staged_event: threading.Event,
+    done_event: threading.Event,
+) -> PendingIOWork:
+    exc_holder: List[BaseException] = []
+
+    def runner() -> None:
+        try:
+            asyncio.run(main())
+        except BaseException as e:  # noqa: B036
+            exc_holder.append(e)
+        finally:
+            stats.end_ts = time.monotonic()
+            staged_event.set()
+            done_event.set()
+
+    thread = threading.Thread(
+        target=runner, name="tsamd-io-pipeline", daemon=True
+    )
+    thread.start()
+    return PendingIOWork(thread, staged_event, done_event, stats, exc_holder)
+
+
+# ---------------------------------------------------------------------------
+# write pipeline
+# ---------------------------------------------------------------------------
+
+
+def execute_write_reqs(
+    write_reqs: List[WriteReq],
+    storage: StoragePlugin,
+    memory_budget_bytes: int,
+    rank: int,
+    is_async: bool = False,
+) -> PendingIOWork:
+    stats = ExecutionStats(total_reqs=len(write_reqs))
+    staged_event = threading.Event()
+    done_event = threading.Event()
+
+    # Largest first: big buffers claim budget early, small ones fill gaps.
+    ordered = sorted(
+        write_reqs, key=lambda r: r.stager.get_staging_cost_bytes(), reverse=True
+    )
+
+    async def main() -> None:
+        budget = _Budget(memory_budget_bytes)
+        io_sem = asyncio.Semaphore(knobs.get_max_io_concurrency())
+        staging_sem = asyncio.Semaphore(knobs.get_num_staging_threads() * 2)
+        executor = ThreadPoolExecutor(
+            max_workers=knobs.get_num_staging_threads(),
+            thread_name_prefix="tsamd-stage",
+        )
+        ctx = StageContext(executor=executor, is_async=is_async)
+        staged_remaining = len(ordered)
+        all_staged = asyncio.Event()
+        if staged_remaining == 0:
+            all_staged.set()
+
+        async def handle(req: WriteReq) -> None:
+            nonlocal staged_remaining
+            cost = req.stager.get_staging_cost_bytes()
+            await budget.acquire(cost)
+            try:
+                async with staging_sem:
+                    buf = await req.stager.stage_buffer(ctx)
+                nbytes = memoryview(buf).nbytes
+                stats.staged_reqs += 1
+                stats.staged_bytes += nbytes
+                staged_remaining -= 1
+                if staged_remaining == 0:
+                    all_staged.set()
+                async with io_sem:
+                    await storage.write(WriteIO(path=req.path, buf=buf))
+                stats.io_bytes += nbytes
+                stats.done_reqs += 1
+                req.stager.release_buffer()
+            finally:
+                await budget.release(cost)
+
+        tasks = [asyncio.create_task(handle(r)) for r in ordered]
+        reporter = asyncio.create_task(_report_progress(stats, rank, "write"))
+
+        async def signal_staged() -> None:
+            await all_staged.wait()
+            stats.staged_ts = time.monotonic()
+            staged_event.set()
+
+        signaler = asyncio.create_task(signal_staged())
+        try:
+            results = await asyncio.gather(*tasks, return_exceptions=True)
+            errors = [r for r in results if isinstance(r, BaseException)]
+            if errors:
+                raise errors[0]
+        finally:
+            signaler.cancel()
+            reporter.cancel()
+            executor.shutdown(wait=False)
+
+    return _spawn_pipeline(main, stats, staged_event, done_event)
+
+
+def sync_execute_write_reqs(
+    write_reqs: List[WriteReq],
+    storage: StoragePlugin,
+    memory_budget_bytes: int,
+    rank: int,
+) -> ExecutionStats:
+    pending = execute_write_reqs(
+        write_reqs, storage, memory_budget_bytes, rank=rank, is_async=False
+    )
+    pending.complete()
+    if rank == 0 and pending.stats.io_bytes:
+        logger.info(
+            "Wrote %.1f MB in %.2fs (%.2f GB/s)",
+            pending.stats.io_bytes / 1e6,
+            (pending.stats.end_ts or 0) - pending.stats.begin_ts,
+            pending.stats.throughput_bytes_per_sec() / 1e9,
+        )
+    return pending.stats
+
+
+# ---------------------------------------------------------------------------
+# read pipeline
+# ---------------------------------------------------------------------------
+
+
+def execute_read_reqs(
+    read_reqs: List[ReadReq],
+    storage: StoragePlugin,
+    memory_budget_bytes: int,
+    rank: int,
+) -> PendingIOWork:
+    stats = ExecutionStats(total_reqs=len(read_reqs))
+    staged_event = threading.Event()
+    done_event = threading.Event()
+
+    ordered = sorted(
+        read_reqs,
+        key=lambda r: r.consumer.get_consuming_cost_bytes(),
+        reverse=True,
+    )
+
+    async def main() -> None:
+        budget = _Budget(memory_budget_bytes)
+        io_sem = asyncio.Semaphore(knobs.get_max_io_concurrency())
+        executor = ThreadPoolExecutor(
+            max_workers=knobs.get_num_staging_threads(),
+            thread_name_prefix="tsamd-consume",
+        )
+        ctx = StageContext(executor=executor)
+
+        async def handle(req: ReadReq) -> None:
+            cost = req.consumer.get_consuming_cost_bytes()
+            await budget.acquire(cost)
+            try:
+                read_io = ReadIO(path=req.path, byte_range=req.byte_range)
+                async with io_sem:
+                    await storage.read(read_io)
+                buf = read_io.buf
+                stats.io_bytes += memoryview(buf).nbytes
+                await req.consumer.consume_buffer(ctx, buf)
+                stats.done_reqs += 1
+            finally:
+                await budget.release(cost)
+
+        tasks = [asyncio.create_task(handle(r)) for r in ordered]
+        reporter = asyncio.create_task(_report_progress(stats, rank, "read"))
+        try:
+            results = await asyncio.gather(*tasks, return_exceptions=True)
+            errors = [r for r in results if isinstance(r, BaseException)]
+            if errors:
+                raise errors[0]
+        finally:
+            reporter.cancel()
+            executor.shutdown(wait=False)
+
+    return _spawn_pipeline(main, stats, staged_event, done_event)
+
+
+def sync_execute_read_reqs(
+    read_reqs: List[ReadReq],
+    storage: StoragePlugin,
+    memory_budget_bytes: int,
+    rank: int,
+) -> ExecutionStats:
+    pending = execute_read_reqs(read_reqs, storage, memory_budget_bytes, rank)
+    pending.complete()
+    if rank == 0 and pending.stats.io_bytes:
+        logger.info(
+            "Read %.1f MB in %.2fs (%.2f GB/s)",
+            pending.stats.io_bytes / 1e6,
+            (pending.stats.end_ts or 0) - pending.stats.begin_ts,
+            pending.stats.throughput_bytes_per_sec() / 1e9,
+        )
+    return pending.stats
+
+
+async def _report_progress(stats: ExecutionStats, rank: int, verb: str) -> None:
+    if rank != 0:
+        return
+    interval = 10.0
+    while True:
+        await asyncio.sleep(interval)
+        logger.info(
+            "[rank %d] %s progress: %d/%d reqs, %.1f MB staged, %.1f MB io",
+            rank,
+            verb,
+            stats.done_reqs,
+            stats.total_reqs,
+            stats.staged_bytes / 1e6,
+            stats.io_bytes / 1e6,
+        )

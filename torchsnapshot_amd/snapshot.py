@@ -1,0 +1,767 @@
+"""Snapshot: the user-facing API.
+
+    Snapshot.take(path, app_state)          — synchronous save
+    Snapshot.async_take(path, app_state)    — non-blocking save (training
+                                              resumes once staging is done)
+    Snapshot(path).restore(app_state)       — in-place load
+    Snapshot(path).read_object(path)        — random access to one object
+    Snapshot(path).get_manifest()           — inspection
+
+Commit protocol: all ranks write payloads, then the ``.snapshot_metadata``
+manifest is written by rank 0 only after a barrier — a snapshot without
+metadata is incomplete by construction (atomicity; parity with reference
+torchsnapshot/snapshot.py:202-209). Async snapshots synchronize the commit
+through a store-based LinearBarrier because collectives must not run on
+the background thread (reference snapshot.py:1010-1032).
+
+Collectives used (all small object collectives over RCCL/gloo): path
+broadcast, replicated-glob intersection, key-set union, partitioner
+assignment, global-manifest gather. Payload bytes never cross ranks.
+"""
+
+from __future__ import annotations
+
+import fnmatch
+import logging
+import threading
+import uuid
+from dataclasses import replace as dataclass_replace
+from typing import Any, Dict, List, Optional, Set, Tuple
+
+import torch
+import torch.distributed as dist
+
+from . import knobs
+from .dist_store import LinearBarrier, get_or_create_store
+from .event import Event, log_event
+from .flatten import Flattened, flatten, inflate
+from .io_preparer import prepare_read, prepare_write
+from .io_types import ReadReq, StoragePlugin, WriteIO, WriteReq
+from .manifest import (
+    ChunkedTensorEntry,
+    Entry,
+    Manifest,
+    METADATA_FILENAME,
+    PrimitiveEntry,
+    SnapshotMetadata,
+    is_container_entry,
+)
+from .manifest_ops import (
+    get_manifest_for_rank,
+    handle_sharded_tensor_elasticity,
+)
+from .manifest_utils import is_fully_replicated_entry
+from .partitioner import PartitionItem, partition_write_reqs
+from .pg_wrapper import PGWrapper
+from .rng_state import RNGState
+from .scheduler import (
+    PendingIOWork,
+    execute_write_reqs,
+    get_process_memory_budget_bytes,
+    sync_execute_read_reqs,
+    sync_execute_write_reqs,
+)
+from .stateful import AppState, Stateful
+from .storage import url_to_storage_plugin
+from .version import __version__
+
+logger = logging.getLogger(__name__)
+
+
+# ---------------------------------------------------------------------------
+# glob matching ("**" crosses path segments, "*" does not)
+# ---------------------------------------------------------------------------
+
+
+def glob_match(path: str, pattern: str) -> bool:
+    return _match_segments(path.split("/"), pattern.split("/"))
+
+
+def _match_segments(segs: List[str], pats: List[str]) -> bool:
+    if not pats:
+        return not segs
+    head, rest = pats[0], pats[1:]
+    if head == "**":
+        # try consuming 0..len(segs) segments
+        for i in range(len(segs) + 1):
+            if _match_segments(segs[i:], rest):
+                return True
+        return False
+    if not segs:
+        return False
+    if not fnmatch.fnmatchcase(segs[0], head):
+        return False
+    return _match_segments(segs[1:], rest)
+
+
+# ---------------------------------------------------------------------------
+# Snapshot
+# ---------------------------------------------------------------------------
+
+
+class Snapshot:
+    def __init__(
+        self,
+        path: str,
+        pg: Optional[dist.ProcessGroup] = None,
+        storage_options: Optional[Dict[str, Any]] = None,
+    ) -> None:
+        self.path = path
+        self.pg = pg
+        self._storage_options = storage_options
+        self._metadata: Optional[SnapshotMetadata] = None
+
+    # -- inspection ---------------------------------------------------------
+
+    @property
+    def metadata(self) -> SnapshotMetadata:
+        if self._metadata is None:
+            storage = url_to_storage_plugin(self.path, self._storage_options)
+            try:
+                from .io_types import ReadIO
+
+                read_io = ReadIO(path=METADATA_FILENAME)
+                try:
+                    storage.sync_read(read_io)
+                except FileNotFoundError:
+                    raise RuntimeError(
+                        f"{self.path} is not a valid snapshot: it is either "
+                        "incomplete or corrupted (missing "
+                        f"{METADATA_FILENAME})"
+                    ) from None
+                self._metadata = SnapshotMetadata.from_str(
+                    bytes(read_io.buf).decode("utf-8")
+                )
+            finally:
+                storage.sync_close()
+        return self._metadata
+
+    def get_manifest(self) -> Dict[str, Any]:
+        """A deep copy of the global manifest as plain dicts."""
+        return {k: v.to_dict() for k, v in self.metadata.manifest.items()}
+
+    # -- save ---------------------------------------------------------------
+
+    @classmethod
+    def take(
+        cls,
+        path: str,
+        app_state: AppState,
+        pg: Optional[dist.ProcessGroup] = None,
+        replicated: Optional[List[str]] = None,
+        storage_options: Optional[Dict[str, Any]] = None,
+    ) -> "Snapshot":
+        torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.take")
+        cls._validate_app_state(app_state)
+        pg_wrapper = PGWrapper(pg)
+        unique_id = uuid.uuid4().hex
+        event_meta = {"id": unique_id, "rank": pg_wrapper.get_rank(), "api": "take"}
+        log_event(Event("take_start", dict(event_meta)))
+        try:
+            path, replicated = cls._coalesce_path_and_replicated(
+                path, app_state, replicated or [], pg_wrapper
+            )
+            storage = url_to_storage_plugin(path, storage_options)
+            try:
+                pending_io_work, metadata = cls._take_impl(
+                    path=path,
+                    app_state=app_state,
+                    storage=storage,
+                    pg_wrapper=pg_wrapper,
+                    replicated=replicated,
+                    is_async=False,
+                )
+                pending_io_work.complete()
+                cls._commit(storage, metadata, pg_wrapper)
+            finally:
+                storage.sync_close()
+            snapshot = cls(path, pg, storage_options)
+            snapshot._metadata = metadata
+            log_event(Event("take_end", {**event_meta, "success": True}))
+            return snapshot
+        except Exception:
+            log_event(Event("take_end", {**event_meta, "success": False}))
+            raise
+
+    @classmethod
+    def async_take(
+        cls,
+        path: str,
+        app_state: AppState,
+        pg: Optional[dist.ProcessGroup] = None,
+        replicated: Optional[List[str]] = None,
+        storage_options: Optional[Dict[str, Any]] = None,
+    ) -> "PendingSnapshot":
+        torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.async_take")
+        cls._validate_app_state(app_state)
+        pg_wrapper = PGWrapper(pg)
+        unique_id = uuid.uuid4().hex
+        event_meta = {
+            "id": unique_id,
+            "rank": pg_wrapper.get_rank(),
+            "api": "async_take",
+        }
+        log_event(Event("async_take_start", dict(event_meta)))
+        path, replicated = cls._coalesce_path_and_replicated(
+            path, app_state, replicated or [], pg_wrapper
+        )
+        storage = url_to_storage_plugin(path, storage_options)
+        # the commit barrier store must be created on the main thread
+        # (bootstrap may use a collective)
+        store = get_or_create_store(pg_wrapper)
+        pending_io_work, metadata = cls._take_impl(
+            path=path,
+            app_state=app_state,
+            storage=storage,
+            pg_wrapper=pg_wrapper,
+            replicated=replicated,
+            is_async=True,
+        )
+        # once staging is complete, the app may mutate its state freely
+        pending_io_work.wait_staged()
+        return PendingSnapshot(
+            path=path,
+            pending_io_work=pending_io_work,
+            pg_wrapper=pg_wrapper,
+            metadata=metadata,
+            storage=storage,
+            store=store,
+            storage_options=storage_options,
+            event_meta=event_meta,
+        )
+
+    @classmethod
+    def _take_impl(
+        cls,
+        path: str,
+        app_state: AppState,
+        storage: StoragePlugin,
+        pg_wrapper: PGWrapper,
+        replicated: List[str],
+        is_async: bool,
+    ) -> Tuple[PendingIOWork, SnapshotMetadata]:
+        rank = pg_wrapper.get_rank()
+        world_size = pg_wrapper.get_world_size()
+
+        # RNG invariance: capture the RNG state before any state_dict()
+        # call can perturb it; restore it afterwards.
+        rng_state_captured = torch.get_rng_state()
+        has_rng_stateful = any(
+            isinstance(v, RNGState) for v in app_state.values()
+        )
+
+        all_keys = cls._gather_keys(app_state, pg_wrapper)
+        manifest: Manifest = {}
+        flattened: Flattened = {}
+        for key in all_keys:
+            if key in app_state:
+                sd = app_state[key].state_dict()
+                m, f = flatten(sd, prefix=key)
+                manifest.update(m)
+                flattened.update(f)
+            # collectives inside state_dict() must not interleave across
+            # statefuls on different ranks
+            pg_wrapper.barrier()
+
+        if has_rng_stateful:
+            torch.set_rng_state(rng_state_captured)
+
+        replicated_paths = cls._calculate_replicated_entries(
+            flattened, replicated, pg_wrapper
+        )
+
+        write_reqs: List[WriteReq] = []
+        req_to_logical: Dict[str, str] = {}
+        for logical_path, obj in flattened.items():
+            entry, reqs = prepare_write(
+                obj=obj,
+                logical_path=logical_path,
+                rank=rank,
+                replicated=logical_path in replicated_paths,
+                is_async_snapshot=is_async,
+            )
+            manifest[logical_path] = entry
+            for r in reqs:
+                req_to_logical[r.path] = logical_path
+            write_reqs.extend(reqs)
+
+        # replicated write-load balancing
+        if world_size > 1 and not knobs.is_partitioner_disabled():
+            write_reqs = cls._partition_replicated(
+                manifest, write_reqs, req_to_logical, replicated_paths,
+                pg_wrapper,
+            )
+
+        write_reqs = _batch(write_reqs)
+
+        global_manifest = cls._gather_manifest(manifest, pg_wrapper)
+        metadata = SnapshotMetadata(
+            version=__version__,
+            world_size=world_size,
+            manifest=global_manifest,
+        )
+
+        budget = get_process_memory_budget_bytes(pg_wrapper)
+        pending = execute_write_reqs(
+            write_reqs, storage, budget, rank=rank, is_async=is_async
+        )
+        return pending, metadata
+
+    @classmethod
+    def _partition_replicated(
+        cls,
+        manifest: Manifest,
+        write_reqs: List[WriteReq],
+        req_to_logical: Dict[str, str],
+        replicated_paths: Set[str],
+        pg_wrapper: PGWrapper,
+    ) -> List[WriteReq]:
+        rank = pg_wrapper.get_rank()
+        items: List[PartitionItem] = []
+        non_replicated_bytes = 0
+        for req in write_reqs:
+            logical = req_to_logical.get(req.path)
+            if logical in replicated_paths and is_fully_replicated_entry(
+                manifest[logical]
+            ):
+                items.append(
+                    PartitionItem(
+                        req_path=req.path,
+                        nbytes=req.stager.get_staging_cost_bytes(),
+                    )
+                )
+            else:
+                non_replicated_bytes += req.stager.get_staging_cost_bytes()
+
+        assignment = partition_write_reqs(items, non_replicated_bytes, pg_wrapper)
+
+        kept_reqs: List[WriteReq] = []
+        dropped_logical: Dict[str, Set[str]] = {}
+        for req in write_reqs:
+            writer = assignment.get(req.path)
+            if writer is None or writer == rank:
+                kept_reqs.append(req)
+            else:
+                logical = req_to_logical[req.path]
+                dropped_logical.setdefault(logical, set()).add(req.path)
+
+        # a rank's manifest only claims replicated payloads it writes
+        for logical, dropped in dropped_logical.items():
+            entry = manifest[logical]
+            if isinstance(entry, ChunkedTensorEntry):
+                kept_chunks = [
+                    c for c in entry.chunks if c.tensor.location not in dropped
+                ]
+                if kept_chunks:
+                    manifest[logical] = dataclass_replace(
+                        entry, chunks=kept_chunks
+                    )
+                else:
+                    del manifest[logical]
+            else:
+                del manifest[logical]
+        return kept_reqs
+
+    @classmethod
+    def _commit(
+        cls,
+        storage: StoragePlugin,
+        metadata: SnapshotMetadata,
+        pg_wrapper: PGWrapper,
+    ) -> None:
+        pg_wrapper.barrier()
+        if pg_wrapper.get_rank() == 0:
+            storage.sync_write(
+                WriteIO(
+                    path=METADATA_FILENAME,
+                    buf=metadata.to_json_str().encode("utf-8"),
+                )
+            )
+        pg_wrapper.barrier()
+
+    # -- load ---------------------------------------------------------------
+
+    def restore(self, app_state: AppState) -> None:
+        torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.restore")
+        self._validate_app_state(app_state)
+        pg_wrapper = PGWrapper(self.pg)
+        event_meta = {
+            "id": uuid.uuid4().hex,
+            "rank": pg_wrapper.get_rank(),
+            "api": "restore",
+        }
+        log_event(Event("restore_start", dict(event_meta)))
+        try:
+            storage = url_to_storage_plugin(self.path, self._storage_options)
+            try:
+                metadata = self.metadata
+                rank_manifest, payload_entries = get_manifest_for_rank(
+                    metadata, pg_wrapper.get_rank()
+                )
+                all_keys = self._gather_keys(app_state, pg_wrapper)
+                # RNG states restore last so nothing after perturbs them
+                all_keys.sort(
+                    key=lambda k: isinstance(app_state.get(k), RNGState)
+                )
+                for key in all_keys:
+                    self._load_stateful(
+                        key=key,
+                        stateful=app_state.get(key),
+                        rank_manifest=rank_manifest,
+                        payload_entries=payload_entries,
+                        storage=storage,
+                        pg_wrapper=pg_wrapper,
+                    )
+                    pg_wrapper.barrier()
+            finally:
+                storage.sync_close()
+            log_event(Event("restore_end", {**event_meta, "success": True}))
+        except Exception:
+            log_event(Event("restore_end", {**event_meta, "success": False}))
+            raise
+
+    def _load_stateful(
+        self,
+        key: str,
+        stateful: Optional[Stateful],
+        rank_manifest: Manifest,
+        payload_entries: Dict[str, Entry],
+        storage: StoragePlugin,
+        pg_wrapper: PGWrapper,
+    ) -> None:
+        if stateful is None:
+            return
+        # in-place targets: load straight into the tensors the stateful
+        # already allocated (halves peak memory)
+        sd = stateful.state_dict()
+        _, flattened_tgt = flatten(sd, prefix=key)
+
+        prefix = f"{key}/"
+        selected = {
+            p: e
+            for p, e in payload_entries.items()
+            if p == key or p.startswith(prefix)
+        }
+        sub_manifest = {
+            p: e
+            for p, e in rank_manifest.items()
+            if p == key or p.startswith(prefix)
+        }
+        if not sub_manifest and not selected:
+            raise RuntimeError(
+                f"no entries found in snapshot for app-state key '{key}'"
+            )
+        handle_sharded_tensor_elasticity(sub_manifest, selected, flattened_tgt)
+
+        read_reqs: List[ReadReq] = []
+        futs: Dict[str, Any] = {}
+        for p, entry in selected.items():
+            obj_out = flattened_tgt.get(p)
+            rr, fut = prepare_read(entry, obj_out)
+            read_reqs.extend(rr)
+            futs[p] = fut
+        read_reqs = _batch_reads(read_reqs)
+        budget = get_process_memory_budget_bytes(pg_wrapper)
+        sync_execute_read_reqs(
+            read_reqs, storage, budget, rank=pg_wrapper.get_rank()
+        )
+        values = {p: f.obj for p, f in futs.items()}
+        state_dict_to_load = inflate(sub_manifest, values, prefix=key)
+        stateful.load_state_dict(state_dict_to_load)
+
+    def get_state_dict_for_key(self, key: str) -> Dict[str, Any]:
+        """Load and return the state dict saved for ``key`` without an
+        in-place target."""
+        pg_wrapper = PGWrapper(self.pg)
+        storage = url_to_storage_plugin(self.path, self._storage_options)
+        try:
+            rank_manifest, payload_entries = get_manifest_for_rank(
+                self.metadata, pg_wrapper.get_rank()
+            )
+            prefix = f"{key}/"
+            selected = {
+                p: e
+                for p, e in payload_entries.items()
+                if p == key or p.startswith(prefix)
+            }
+            sub_manifest = {
+                p: e
+                for p, e in rank_manifest.items()
+                if p == key or p.startswith(prefix)
+            }
+            if not sub_manifest and not selected:
+                raise RuntimeError(f"key '{key}' not found in snapshot")
+            read_reqs: List[ReadReq] = []
+            futs: Dict[str, Any] = {}
+            for p, entry in selected.items():
+                rr, fut = prepare_read(entry, None)
+                read_reqs.extend(rr)
+                futs[p] = fut
+            read_reqs = _batch_reads(read_reqs)
+            budget = get_process_memory_budget_bytes(pg_wrapper)
+            sync_execute_read_reqs(
+                read_reqs, storage, budget, rank=pg_wrapper.get_rank()
+            )
+            values = {p: f.obj for p, f in futs.items()}
+            return inflate(sub_manifest, values, prefix=key)
+        finally:
+            storage.sync_close()
+
+    def read_object(
+        self,
+        path: str,
+        obj_out: Optional[Any] = None,
+        memory_budget_bytes: Optional[int] = None,
+    ) -> Any:
+        """Random access: load one object by its global manifest path
+        ("<rank>/<logical_path>"). ``memory_budget_bytes`` bounds peak host
+        memory via tiled byte-range reads."""
+        torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.read_object")
+        rank_str, _, logical_path = path.partition("/")
+        try:
+            rank = int(rank_str)
+        except ValueError:
+            raise ValueError(
+                f"read_object path must start with a rank, got {path!r}"
+            ) from None
+        _, payload_entries = get_manifest_for_rank(self.metadata, rank)
+        if logical_path not in payload_entries:
+            raise ValueError(
+                f"path '{path}' does not exist in the snapshot (no entry "
+                f"'{logical_path}' for rank {rank})"
+            )
+        entry = payload_entries[logical_path]
+        if isinstance(entry, PrimitiveEntry):
+            return entry.get_value()
+        storage = url_to_storage_plugin(self.path, self._storage_options)
+        try:
+            read_reqs, fut = prepare_read(
+                entry, obj_out, buffer_size_limit_bytes=memory_budget_bytes
+            )
+            read_reqs = _batch_reads(read_reqs)
+            budget = memory_budget_bytes or (32 * 1024**3)
+            sync_execute_read_reqs(read_reqs, storage, budget, rank=0)
+            return fut.obj
+        finally:
+            storage.sync_close()
+
+    # -- shared helpers -----------------------------------------------------
+
+    @staticmethod
+    def _validate_app_state(app_state: AppState) -> None:
+        if not isinstance(app_state, dict):
+            raise TypeError(
+                f"app_state must be Dict[str, Stateful], got {type(app_state)}"
+            )
+        for key, value in app_state.items():
+            if not isinstance(value, Stateful):
+                raise TypeError(
+                    f"app_state['{key}'] ({type(value).__name__}) does not "
+                    "implement state_dict()/load_state_dict()"
+                )
+
+    @classmethod
+    def _coalesce_path_and_replicated(
+        cls,
+        path: str,
+        app_state: AppState,
+        replicated: List[str],
+        pg_wrapper: PGWrapper,
+    ) -> Tuple[str, List[str]]:
+        # all ranks must agree on the snapshot path: rank 0 wins
+        obj_list = [path]
+        pg_wrapper.broadcast_object_list(obj_list, src=0)
+        if obj_list[0] != path:
+            logger.warning(
+                "rank %d: snapshot path %r differs from rank 0's %r; using "
+                "rank 0's",
+                pg_wrapper.get_rank(),
+                path,
+                obj_list[0],
+            )
+        path = obj_list[0]
+
+        replicated = list(replicated) + cls._infer_replicated(app_state)
+        # a pattern counts only if every rank requested it
+        gathered: List[Optional[List[str]]] = [None] * pg_wrapper.get_world_size()
+        pg_wrapper.all_gather_object(gathered, sorted(set(replicated)))
+        common = set(gathered[0] or [])
+        for lst in gathered[1:]:
+            common &= set(lst or [])
+        dropped = set(replicated) - common
+        if dropped:
+            logger.warning(
+                "replicated patterns %s were not requested on all ranks; "
+                "ignoring them",
+                sorted(dropped),
+            )
+        return path, sorted(common)
+
+    @staticmethod
+    def _infer_replicated(app_state: AppState) -> List[str]:
+        """DDP modules are replicated by construction: mark their whole
+        subtree (minus parameters_to_ignore)."""
+        patterns: List[str] = []
+        try:
+            from torch.nn.parallel import DistributedDataParallel as DDP
+        except ImportError:
+            return patterns
+        for key, stateful in app_state.items():
+            if not isinstance(stateful, DDP):
+                continue
+            ignored = set(
+                getattr(stateful, "parameters_to_ignore", None) or []
+            )
+            if not ignored:
+                patterns.append(f"{key}/**")
+            else:
+                for name in stateful.state_dict().keys():
+                    if name not in ignored:
+                        patterns.append(f"{key}/{name}")
+        return patterns
+
+    @classmethod
+    def _calculate_replicated_entries(
+        cls,
+        flattened: Flattened,
+        replicated: List[str],
+        pg_wrapper: PGWrapper,
+    ) -> Set[str]:
+        if not replicated:
+            return set()
+        candidates = sorted(
+            p
+            for p in flattened.keys()
+            if any(glob_match(p, pat) for pat in replicated)
+        )
+        # a path is replicated only if it matched on EVERY rank
+        gathered: List[Optional[List[str]]] = [None] * pg_wrapper.get_world_size()
+        pg_wrapper.all_gather_object(gathered, candidates)
+        common = set(gathered[0] or [])
+        for lst in gathered[1:]:
+            common &= set(lst or [])
+        return common
+
+    @staticmethod
+    def _gather_keys(app_state: AppState, pg_wrapper: PGWrapper) -> List[str]:
+        gathered: List[Optional[List[str]]] = [None] * pg_wrapper.get_world_size()
+        pg_wrapper.all_gather_object(gathered, sorted(app_state.keys()))
+        union: Set[str] = set()
+        for lst in gathered:
+            union |= set(lst or [])
+        return sorted(union)
+
+    @staticmethod
+    def _gather_manifest(manifest: Manifest, pg_wrapper: PGWrapper) -> Manifest:
+        local = {p: e.to_dict() for p, e in manifest.items()}
+        gathered: List[Optional[Dict[str, Dict[str, Any]]]] = [
+            None
+        ] * pg_wrapper.get_world_size()
+        pg_wrapper.all_gather_object(gathered, local)
+        from .manifest import entry_from_dict
+
+        global_manifest: Manifest = {}
+        for rank, rank_manifest in enumerate(gathered):
+            for p, d in (rank_manifest or {}).items():
+                global_manifest[f"{rank}/{p}"] = entry_from_dict(d)
+        return global_manifest
+
+
+def _batch(write_reqs: List[WriteReq]) -> List[WriteReq]:
+    from .batcher import batch_write_requests
+
+    return batch_write_requests(write_reqs)
+
+
+def _batch_reads(read_reqs: List[ReadReq]) -> List[ReadReq]:
+    from .batcher import batch_read_requests
+
+    return batch_read_requests(read_reqs)
+
+
+# ---------------------------------------------------------------------------
+# PendingSnapshot
+# ---------------------------------------------------------------------------
+
+
+class PendingSnapshot:
+    """Returned by async_take: storage I/O drains on a background thread;
+    the commit is coordinated with a store barrier (no collectives off the
+    main thread)."""
+
+    def __init__(
+        self,
+        path: str,
+        pending_io_work: PendingIOWork,
+        pg_wrapper: PGWrapper,
+        metadata: SnapshotMetadata,
+        storage: StoragePlugin,
+        store,
+        storage_options: Optional[Dict[str, Any]],
+        event_meta: Optional[Dict[str, Any]] = None,
+    ) -> None:
+        self.path = path
+        self._pending_io_work = pending_io_work
+        self._pg_wrapper = pg_wrapper
+        self._metadata = metadata
+        self._storage = storage
+        self._storage_options = storage_options
+        self._event_meta = event_meta or {}
+        self._barrier = LinearBarrier(
+            prefix=f"tsamd_commit_{path}",
+            store=store,
+            rank=pg_wrapper.get_rank(),
+            world_size=pg_wrapper.get_world_size(),
+        )
+        self._exc: Optional[BaseException] = None
+        self._done_event = threading.Event()
+        self._thread = threading.Thread(
+            target=self._complete, name="tsamd-async-commit", daemon=True
+        )
+        self._thread.start()
+
+    def _complete(self) -> None:
+        try:
+            self._pending_io_work.complete()
+            self._barrier.arrive()
+            if self._pg_wrapper.get_rank() == 0:
+                self._storage.sync_write(
+                    WriteIO(
+                        path=METADATA_FILENAME,
+                        buf=self._metadata.to_json_str().encode("utf-8"),
+                    )
+                )
+            self._barrier.depart()
+            log_event(
+                Event("async_take_end", {**self._event_meta, "success": True})
+            )
+        except BaseException as e:  # noqa: B036
+            self._exc = e
+            try:
+                self._barrier.report_error(e)
+            except Exception:
+                logger.exception("failed to report async-commit error")
+            log_event(
+                Event("async_take_end", {**self._event_meta, "success": False})
+            )
+        finally:
+            try:
+                self._storage.sync_close()
+            except Exception:
+                logger.exception("failed to close storage after async take")
+            self._done_event.set()
+
+    def wait(self) -> Snapshot:
+        self._thread.join()
+        if self._exc is not None:
+            raise RuntimeError(
+                "async snapshot failed; no metadata was committed"
+            ) from self._exc
+        snapshot = Snapshot(
+            self.path, self._pg_wrapper.pg, self._storage_options
+        )
+        snapshot._metadata = self._metadata
+        return snapshot
+
+    def done(self) -> bool:
+        return self._done_event.is_set()
