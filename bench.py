@@ -1,0 +1,221 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: checkpoint-save throughput for an FSDP-sharded
+Llama-3-8B (bf16, random-init, synthetic) written to local NVMe.
+
+One "step" = one full Snapshot.take of the model state. The whole-job
+metric is aggregate save GB/s (model bytes / wall time per save, maxed
+over ranks). Also reports the async_take stall time (how long training is
+blocked by a non-blocking snapshot).
+
+Launch (driver contract):
+    python bench.py --gpus 1 --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+At N>1 each rank holds 1/N of every parameter as a DTensor on a 1-D mesh
+(FSDP2-style layout); at N=1 parameters are plain device tensors.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import shutil
+import time
+from typing import Dict, List, Tuple
+
+import torch
+
+# Reference baseline: implied GB/s of torchsnapshot's published 20 GB DDP
+# save to local FS on p4d.24xlarge (1 GPU ~13.91 s, 8 GPUs
+# ~3.38 s). 2/4-GPU points interpolated geometrically.
+_BASELINE_GBPS = {1: 1.44, 2: 2.29, 4: 3.65, 8: 5.92}
+
+
+def llama3_8b_shapes() -> List[Tuple[str, Tuple[int, ...]]]:
+    """Parameter shapes of Llama-3-8B (vocab 128256, hidden 4096,
+    intermediate 14336, 32 layers, 32 heads / 8 KV heads)."""
+    V, H, I, L = 128256, 4096, 14336, 32
+    KV = 1024  # 8 kv heads * 128 head_dim
+    shapes: List[Tuple[str, Tuple[int, ...]]] = [
+        ("model.embed_tokens.weight", (V, H)),
+        ("model.norm.weight", (H,)),
+        ("lm_head.weight", (V, H)),
+    ]
+    for i in range(L):
+        p = f"model.layers.{i}"
+        shapes += [
+            (f"{p}.self_attn.q_proj.weight", (H, H)),
+            (f"{p}.self_attn.k_proj.weight", (KV, H)),
+            (f"{p}.self_attn.v_proj.weight", (KV, H)),
+            (f"{p}.self_attn.o_proj.weight", (H, H)),
+            (f"{p}.mlp.gate_proj.weight", (I, H)),
+            (f"{p}.mlp.up_proj.weight", (I, H)),
+            (f"{p}.mlp.down_proj.weight", (H, I)),
+            (f"{p}.input_layernorm.weight", (H,)),
+            (f"{p}.post_attention_layernorm.weight", (H,)),
+        ]
+    return shapes
+
+
+class _BenchState:
+    """Stateful wrapper over the synthetic model state dict."""
+
+    def __init__(self, sd: Dict[str, torch.Tensor]) -> None:
+        self._sd = sd
+
+    def state_dict(self) -> Dict[str, torch.Tensor]:
+        return self._sd
+
+    def load_state_dict(self, sd: Dict[str, torch.Tensor]) -> None:
+        self._sd = sd
+
+
+def build_state(
+    device: torch.device, world_size: int, dtype: torch.dtype
+) -> Tuple[_BenchState, int]:
+    shapes = llama3_8b_shapes()
+    total_bytes = 0
+    sd: Dict[str, torch.Tensor] = {}
+    if world_size > 1:
+        from torch.distributed.device_mesh import init_device_mesh
+        from torch.distributed.tensor import DTensor
+        from torch.distributed.tensor.placement_types import Shard
+
+        mesh = init_device_mesh("cuda", (world_size,))
+        for name, shape in shapes:
+            assert shape[0] % world_size == 0, (name, shape)
+            local_shape = (shape[0] // world_size,) + tuple(shape[1:])
+            local = torch.empty(local_shape, dtype=dtype, device=device)
+            local.normal_(0, 0.02)
+            sd[name] = DTensor.from_local(local, mesh, [Shard(0)])
+            total_bytes += int(
+                torch.Size(shape).numel() * local.element_size()
+            )
+    else:
+        for name, shape in shapes:
+            t = torch.empty(shape, dtype=dtype, device=device)
+            t.normal_(0, 0.02)
+            sd[name] = t
+            total_bytes += t.numel() * t.element_size()
+    return _BenchState(sd), total_bytes
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=3)
+    parser.add_argument("--warmup", type=int, default=1)
+    parser.add_argument(
+        "--dir", type=str, default=os.environ.get("TSAMD_BENCH_DIR", "")
+    )
+    parser.add_argument(
+        "--keep", action="store_true", help="keep the checkpoint directory"
+    )
+    args = parser.parse_args()
+
+    from torchsnapshot_amd import Snapshot
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world_size > 1:
+        import torch.distributed as dist
+
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl")
+    device = torch.device("cuda", local_rank)
+    torch.cuda.set_device(device)
+
+    bench_dir = args.dir or "/tmp/tsamd_bench"
+    ckpt_path = os.path.join(bench_dir, "ckpt")
+    if rank == 0:
+        os.makedirs(bench_dir, exist_ok=True)
+        shutil.rmtree(ckpt_path, ignore_errors=True)
+    _barrier(world_size)
+
+    state, total_bytes = build_state(device, world_size, torch.bfloat16)
+    app_state = {"model": state}
+
+    # warmup (untimed): allocates pinned blocks, compiles nothing, warms
+    # the fs dir cache
+    for _ in range(args.warmup):
+        Snapshot.take(ckpt_path, app_state)
+    _barrier(world_size)
+    torch.cuda.synchronize(device)
+
+    t0 = time.monotonic()
+    for _ in range(args.steps):
+        Snapshot.take(ckpt_path, app_state)
+    torch.cuda.synchronize(device)
+    _barrier(world_size)
+    elapsed = time.monotonic() - t0
+    elapsed = _max_over_ranks(elapsed, world_size, device)
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    gbps = (total_bytes / 1e9) / (elapsed / args.steps)
+
+    # async_take stall: how long the "training thread" is blocked
+    _barrier(world_size)
+    t0 = time.monotonic()
+    pending = Snapshot.async_take(ckpt_path, app_state)
+    stall_s = time.monotonic() - t0
+    pending.wait()
+    stall_s = _max_over_ranks(stall_s, world_size, device)
+
+    if rank == 0 and not args.keep:
+        shutil.rmtree(ckpt_path, ignore_errors=True)
+
+    if rank == 0:
+        baseline = _BASELINE_GBPS.get(world_size)
+        result = {
+            "metric": "checkpoint_save_GBps",
+            "value": round(gbps, 3),
+            "unit": "GB/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 1),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": round(gbps / baseline, 3) if baseline else None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "stall_sec": round(stall_s, 3),
+            "config": {
+                "model": "llama-3-8b",
+                "model_bytes": total_bytes,
+                "global_batch": None,
+                "seq_len": None,
+                "parallelism": f"fsdp{world_size}" if world_size > 1 else "single",
+                "storage": bench_dir,
+            },
+        }
+        print(json.dumps(result))
+
+    if world_size > 1:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+def _barrier(world_size: int) -> None:
+    if world_size > 1:
+        import torch.distributed as dist
+
+        dist.barrier()
+
+
+def _max_over_ranks(value: float, world_size: int, device) -> float:
+    if world_size <= 1:
+        return value
+    import torch.distributed as dist
+
+    t = torch.tensor([value], dtype=torch.float64, device=device)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
+
+
+if __name__ == "__main__":
+    main()

@@ -17,7 +17,6 @@ consumer (parity with torchsnapshot/batcher.py:358-478).
 from __future__ import annotations
 
 import asyncio
-import uuid
 from collections import defaultdict
 from typing import Dict, List, Sequence, Tuple
 
@@ -48,9 +47,13 @@ def _is_batchable(req: WriteReq) -> bool:
     )
 
 
-def batch_write_requests(write_reqs: List[WriteReq]) -> List[WriteReq]:
+def batch_write_requests(
+    write_reqs: List[WriteReq], rank: int = 0
+) -> List[WriteReq]:
     """Group batchable requests into slab writes; returns the new request
-    list (member entries are relocated in place)."""
+    list (member entries are relocated in place). Slab names are
+    deterministic per rank within one snapshot, so re-saving to the same
+    path overwrites instead of accumulating slabs."""
     if knobs.is_batching_disabled():
         return write_reqs
     out: List[WriteReq] = []
@@ -62,6 +65,7 @@ def batch_write_requests(write_reqs: List[WriteReq]) -> List[WriteReq]:
             out.append(req)
 
     slab_limit = knobs.get_slab_size_threshold_bytes()
+    seq = 0
     for device_str, members in groups.items():
         if len(members) == 1:
             out.extend(members)
@@ -73,19 +77,23 @@ def batch_write_requests(write_reqs: List[WriteReq]) -> List[WriteReq]:
             nbytes = req.stager.tensor.numel() * req.stager.tensor.element_size()
             aligned = (nbytes + ALIGN - 1) // ALIGN * ALIGN
             if cur and cur_bytes + aligned > slab_limit:
-                out.append(_make_slab(device_str, cur))
+                out.append(_make_slab(device_str, cur, rank, seq))
+                seq += 1
                 cur, cur_bytes = [], 0
             cur.append(req)
             cur_bytes += aligned
         if len(cur) == 1:
             out.append(cur[0])
         elif cur:
-            out.append(_make_slab(device_str, cur))
+            out.append(_make_slab(device_str, cur, rank, seq))
+            seq += 1
     return out
 
 
-def _make_slab(device_str: str, members: List[WriteReq]) -> WriteReq:
-    slab_path = f"batched/{uuid.uuid4().hex}"
+def _make_slab(
+    device_str: str, members: List[WriteReq], rank: int, seq: int
+) -> WriteReq:
+    slab_path = f"batched/{rank}-{seq}"
     tensors = [m.stager.tensor.detach() for m in members]
     _, offsets, total = build_pack_items(tensors)
     for m, off in zip(members, offsets):

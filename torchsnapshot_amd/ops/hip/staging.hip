@@ -1,0 +1,470 @@
+// MI355X (gfx950) staging data plane for torchsnapshot_amd.
+//
+// What lives here:
+//   - per-device side HIP streams: checkpoint D2H traffic runs beside the
+//     training step's streams, so staging overlaps compute
+//   - gather-pack kernel: N strided device tensors -> one contiguous slab,
+//     in a single launch (replaces the reference's per-tensor
+//     UntypedStorage copies + ByteTensor slab, torchsnapshot/batcher.py
+//     :104-162, and its tensor.cpu() staging, io_preparers/tensor.py:353)
+//   - scatter kernel (same code path, direction-reversed) for restore
+//   - SDMA copies (hipMemcpyAsync) slab <-> pinned host memory; the slab
+//     can also be skipped entirely ("direct" mode): the kernel writes
+//     gathered bytes straight into pinned host memory over PCIe
+//
+// Design notes (CDNA4):
+//   - wave64; block = 256 threads; work units of 64 KiB of slab bytes,
+//     grid-strided with >> 256 workgroups so all 8 XCDs fill
+//   - wide rows (>= 2 KiB contiguous) copy lane-across-columns with the
+//     widest vector the layout allows (16 B/lane -> 1 KiB per wave per
+//     instruction); narrow rows copy lane-per-row so writes to the slab
+//     stay coalesced across lanes even when reads are scattered
+//   - all within-tensor indices fit u32 (tensors are chunked to <= 512 MB
+//     upstream); row->coordinate decomposition uses u32 div/mod
+//
+// Deliberately torch-free: tensors enter as raw pointers; stream ordering
+// against the producer is a caller-provided stream handle we event-chain.
+// This keeps the extension ABI-independent of the torch build.
+
+#include <hip/hip_runtime.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstdint>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess) {                                                 \
+      throw std::runtime_error(std::string("HIP error at " #expr ": ") +    \
+                               hipGetErrorString(_e));                      \
+    }                                                                       \
+  } while (0)
+
+namespace {
+
+constexpr int kMaxDims = 6;
+constexpr uint32_t kWorkUnitBytes = 64 * 1024;
+constexpr int kBlockThreads = 256;
+constexpr uint32_t kWideRowBytes = 2048;
+
+struct Desc {
+  const char* tensor_base;
+  unsigned long long flat_off;   // byte offset of this tensor in the slab
+  uint32_t nbytes;               // payload bytes (< 2^32 by construction)
+  uint32_t row_bytes;            // innermost contiguous run
+  uint32_t vec;                  // safe vector width (16/8/4/2/1)
+  uint32_t ndim;                 // number of outer (strided) dims
+  uint32_t sizes[kMaxDims];
+  long long strides[kMaxDims];   // byte strides of outer dims
+};
+
+// byte offset of row `row` within the strided tensor (excluding columns)
+__device__ inline unsigned long long row_offset(const Desc& d, uint32_t row) {
+  unsigned long long off = 0;
+  uint32_t rem = row;
+#pragma unroll
+  for (int k = kMaxDims - 1; k >= 0; --k) {
+    if (k >= (int)d.ndim) continue;
+    uint32_t sz = d.sizes[k];
+    uint32_t idx = rem % sz;
+    rem /= sz;
+    off += (unsigned long long)idx * (unsigned long long)d.strides[k];
+  }
+  return off;
+}
+
+template <int VEC, bool GATHER>
+__device__ inline void copy_vec(char* flat, const char* strided) {
+  if (GATHER) {
+    if (VEC == 16)
+      *reinterpret_cast<uint4*>(flat) = *reinterpret_cast<const uint4*>(strided);
+    else if (VEC == 8)
+      *reinterpret_cast<uint64_t*>(flat) = *reinterpret_cast<const uint64_t*>(strided);
+    else if (VEC == 4)
+      *reinterpret_cast<uint32_t*>(flat) = *reinterpret_cast<const uint32_t*>(strided);
+    else if (VEC == 2)
+      *reinterpret_cast<uint16_t*>(flat) = *reinterpret_cast<const uint16_t*>(strided);
+    else
+      *flat = *strided;
+  } else {
+    if (VEC == 16)
+      *reinterpret_cast<uint4*>(const_cast<char*>(strided)) = *reinterpret_cast<const uint4*>(flat);
+    else if (VEC == 8)
+      *reinterpret_cast<uint64_t*>(const_cast<char*>(strided)) = *reinterpret_cast<const uint64_t*>(flat);
+    else if (VEC == 4)
+      *reinterpret_cast<uint32_t*>(const_cast<char*>(strided)) = *reinterpret_cast<const uint32_t*>(flat);
+    else if (VEC == 2)
+      *reinterpret_cast<uint16_t*>(const_cast<char*>(strided)) = *reinterpret_cast<const uint16_t*>(flat);
+    else
+      *const_cast<char*>(strided) = *flat;
+  }
+}
+
+// Copy [s, e) of the tensor's flat byte range for one work unit.
+template <int VEC, bool GATHER>
+__device__ void process_range(const Desc& d, char* flat_base, uint32_t s,
+                              uint32_t e) {
+  char* flat = flat_base + d.flat_off;
+  const uint32_t tid = threadIdx.x;
+  const uint32_t rb = d.row_bytes;
+  if (d.ndim == 0 || rb == d.nbytes) {
+    // fully contiguous: plain vectorized copy of [s, e)
+    const char* src = d.tensor_base;
+    for (uint32_t i = s + tid * VEC; i < e; i += kBlockThreads * VEC) {
+      copy_vec<VEC, GATHER>(flat + i, src + i);
+    }
+    return;
+  }
+  if (rb >= kWideRowBytes) {
+    // wide rows: rows sequential, lanes across columns
+    uint32_t row = s / rb;
+    uint32_t col = s - row * rb;
+    uint32_t off = s;
+    while (off < e) {
+      const char* src = d.tensor_base + row_offset(d, row);
+      uint32_t n = min(rb - col, e - off);
+      for (uint32_t i = tid * VEC; i < n; i += kBlockThreads * VEC) {
+        copy_vec<VEC, GATHER>(flat + off + i, src + col + i);
+      }
+      off += n;
+      ++row;
+      col = 0;
+    }
+  } else {
+    // narrow rows: one lane per row; slab-side access stays coalesced
+    uint32_t first_row = s / rb;
+    uint32_t last_row = (e - 1) / rb;
+    for (uint32_t r = first_row + tid; r <= last_row; r += kBlockThreads) {
+      const char* src = d.tensor_base + row_offset(d, r);
+      uint32_t flat_pos = r * rb;
+      uint32_t lo = flat_pos < s ? (s - flat_pos) : 0;
+      uint32_t hi = (flat_pos + rb > e) ? (e - flat_pos) : rb;
+      for (uint32_t i = lo; i < hi; i += VEC) {
+        copy_vec<VEC, GATHER>(flat + flat_pos + i, src + i);
+      }
+    }
+  }
+}
+
+template <bool GATHER>
+__global__ __launch_bounds__(kBlockThreads) void pack_kernel(
+    const Desc* __restrict__ descs, int n,
+    const unsigned long long* __restrict__ wu_prefix,
+    unsigned long long total_wus, char* __restrict__ flat_base) {
+  for (unsigned long long wu = blockIdx.x; wu < total_wus; wu += gridDim.x) {
+    // binary search: which tensor owns this work unit
+    int lo = 0, hi = n - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (wu_prefix[mid] <= wu) lo = mid;
+      else hi = mid - 1;
+    }
+    const Desc d = descs[lo];
+    unsigned long long local_wu = wu - wu_prefix[lo];
+    uint32_t s = (uint32_t)(local_wu * kWorkUnitBytes);
+    uint32_t e = s + kWorkUnitBytes > d.nbytes ? d.nbytes
+                                               : s + kWorkUnitBytes;
+    switch (d.vec) {
+      case 16: process_range<16, GATHER>(d, flat_base, s, e); break;
+      case 8: process_range<8, GATHER>(d, flat_base, s, e); break;
+      case 4: process_range<4, GATHER>(d, flat_base, s, e); break;
+      case 2: process_range<2, GATHER>(d, flat_base, s, e); break;
+      default: process_range<1, GATHER>(d, flat_base, s, e); break;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host-side op management
+// ---------------------------------------------------------------------------
+
+struct DeviceCtx {
+  hipStream_t d2h_stream = nullptr;
+  hipStream_t h2d_stream = nullptr;
+};
+
+struct Op {
+  hipEvent_t ev = nullptr;
+  int device = 0;
+  void* scratch = nullptr;  // freed at wait()
+  std::vector<char> host_staging;
+};
+
+std::mutex g_mu;
+std::unordered_map<int, DeviceCtx> g_ctx;
+std::unordered_map<long long, Op> g_ops;
+long long g_next_handle = 1;
+
+DeviceCtx& get_ctx(int device) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  auto it = g_ctx.find(device);
+  if (it == g_ctx.end()) {
+    HIP_CHECK(hipSetDevice(device));
+    DeviceCtx ctx;
+    HIP_CHECK(hipStreamCreateWithFlags(&ctx.d2h_stream, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&ctx.h2d_stream, hipStreamNonBlocking));
+    it = g_ctx.emplace(device, ctx).first;
+  }
+  return it->second;
+}
+
+long long register_op(hipEvent_t ev, int device, void* scratch,
+                      std::vector<char>&& staging) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  long long h = g_next_handle++;
+  Op op;
+  op.ev = ev;
+  op.device = device;
+  op.scratch = scratch;
+  op.host_staging = std::move(staging);
+  g_ops.emplace(h, std::move(op));
+  return h;
+}
+
+void chain_after(hipStream_t side, uintptr_t producer_stream, int device) {
+  // make the side stream wait for work already enqueued on the producer's
+  // stream (the training step that materialized the tensors)
+  hipEvent_t dep;
+  HIP_CHECK(hipEventCreateWithFlags(&dep, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(dep, reinterpret_cast<hipStream_t>(producer_stream)));
+  HIP_CHECK(hipStreamWaitEvent(side, dep, 0));
+  HIP_CHECK(hipEventDestroy(dep));
+}
+
+// flat desc rows from python: 18 u64 each (see ops/staging.py)
+constexpr int kRowInts = 18;
+
+std::vector<Desc> parse_descs(const std::vector<unsigned long long>& flat,
+                              int n) {
+  if ((int)flat.size() != n * kRowInts) {
+    throw std::runtime_error("bad descriptor array length");
+  }
+  std::vector<Desc> descs(n);
+  for (int i = 0; i < n; ++i) {
+    const unsigned long long* r = flat.data() + i * kRowInts;
+    Desc& d = descs[i];
+    d.tensor_base = reinterpret_cast<const char*>(r[0]);
+    d.flat_off = r[1];
+    if (r[2] >= (1ull << 32)) {
+      throw std::runtime_error(
+          "tensor too large for pack kernel (>=4GB); chunk it upstream");
+    }
+    d.nbytes = (uint32_t)r[2];
+    d.row_bytes = (uint32_t)r[3];
+    d.vec = (uint32_t)r[4];
+    d.ndim = (uint32_t)r[5];
+    for (int k = 0; k < kMaxDims; ++k) {
+      d.sizes[k] = (uint32_t)r[6 + k];
+      d.strides[k] = (long long)r[12 + k];
+    }
+    if (d.ndim > (uint32_t)kMaxDims) {
+      throw std::runtime_error("too many outer dims");
+    }
+  }
+  return descs;
+}
+
+long long launch_pack(const std::vector<unsigned long long>& flat, int n,
+                      uintptr_t slab_ptr, uintptr_t pinned_ptr,
+                      unsigned long long total_bytes, uintptr_t producer_stream,
+                      int device, bool gather) {
+  HIP_CHECK(hipSetDevice(device));
+  DeviceCtx& ctx = get_ctx(device);
+  hipStream_t stream = gather ? ctx.d2h_stream : ctx.h2d_stream;
+
+  std::vector<Desc> descs = parse_descs(flat, n);
+  std::vector<unsigned long long> wu_prefix(n + 1, 0);
+  for (int i = 0; i < n; ++i) {
+    unsigned long long wus =
+        (descs[i].nbytes + kWorkUnitBytes - 1) / kWorkUnitBytes;
+    wu_prefix[i + 1] = wu_prefix[i] + wus;
+  }
+  unsigned long long total_wus = wu_prefix[n];
+
+  // flat side: device slab if provided, else the pinned host buffer
+  // (device-addressable) for direct PCIe writes/reads
+  char* flat_base;
+  if (slab_ptr != 0) {
+    flat_base = reinterpret_cast<char*>(slab_ptr);
+  } else {
+    void* dev_ptr = nullptr;
+    hipError_t err = hipHostGetDevicePointer(
+        &dev_ptr, reinterpret_cast<void*>(pinned_ptr), 0);
+    flat_base = reinterpret_cast<char*>(
+        err == hipSuccess && dev_ptr ? dev_ptr
+                                     : reinterpret_cast<void*>(pinned_ptr));
+  }
+
+  chain_after(stream, producer_stream, device);
+
+  // stage descriptors + prefix sums in one device allocation
+  size_t desc_bytes = sizeof(Desc) * n;
+  size_t prefix_bytes = sizeof(unsigned long long) * (n + 1);
+  std::vector<char> staging(desc_bytes + prefix_bytes);
+  memcpy(staging.data(), descs.data(), desc_bytes);
+  memcpy(staging.data() + desc_bytes, wu_prefix.data(), prefix_bytes);
+  void* scratch = nullptr;
+  HIP_CHECK(hipMallocAsync(&scratch, staging.size(), stream));
+  HIP_CHECK(hipMemcpyAsync(scratch, staging.data(), staging.size(),
+                           hipMemcpyHostToDevice, stream));
+
+  const Desc* d_descs = reinterpret_cast<const Desc*>(scratch);
+  const unsigned long long* d_prefix =
+      reinterpret_cast<const unsigned long long*>(
+          reinterpret_cast<char*>(scratch) + desc_bytes);
+
+  if (!gather && slab_ptr != 0) {
+    // restore via slab: H2D copy first, then scatter out of the slab
+    HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(slab_ptr),
+                             reinterpret_cast<void*>(pinned_ptr), total_bytes,
+                             hipMemcpyHostToDevice, stream));
+  }
+
+  unsigned int grid = (unsigned int)std::min<unsigned long long>(
+      total_wus == 0 ? 1 : total_wus, 16384ull);
+  if (gather) {
+    hipLaunchKernelGGL(pack_kernel<true>, dim3(grid), dim3(kBlockThreads), 0,
+                       stream, d_descs, n, d_prefix, total_wus, flat_base);
+  } else {
+    hipLaunchKernelGGL(pack_kernel<false>, dim3(grid), dim3(kBlockThreads), 0,
+                       stream, d_descs, n, d_prefix, total_wus, flat_base);
+  }
+  HIP_CHECK(hipGetLastError());
+
+  if (gather && slab_ptr != 0) {
+    // slab mode: one SDMA copy moves the packed slab to pinned host memory
+    HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(pinned_ptr),
+                             reinterpret_cast<void*>(slab_ptr), total_bytes,
+                             hipMemcpyDeviceToHost, stream));
+  }
+  HIP_CHECK(hipFreeAsync(scratch, stream));
+
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(ev, stream));
+  return register_op(ev, device, nullptr, std::move(staging));
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// python API
+// ---------------------------------------------------------------------------
+
+static long long d2h_copy(uintptr_t src, uintptr_t dst_pinned,
+                          unsigned long long nbytes, uintptr_t producer_stream,
+                          int device) {
+  HIP_CHECK(hipSetDevice(device));
+  DeviceCtx& ctx = get_ctx(device);
+  chain_after(ctx.d2h_stream, producer_stream, device);
+  HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dst_pinned),
+                           reinterpret_cast<void*>(src), nbytes,
+                           hipMemcpyDeviceToHost, ctx.d2h_stream));
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(ev, ctx.d2h_stream));
+  return register_op(ev, device, nullptr, {});
+}
+
+static long long h2d_copy(uintptr_t src_pinned, uintptr_t dst,
+                          unsigned long long nbytes, uintptr_t producer_stream,
+                          int device) {
+  HIP_CHECK(hipSetDevice(device));
+  DeviceCtx& ctx = get_ctx(device);
+  chain_after(ctx.h2d_stream, producer_stream, device);
+  HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dst),
+                           reinterpret_cast<void*>(src_pinned), nbytes,
+                           hipMemcpyHostToDevice, ctx.h2d_stream));
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(ev, ctx.h2d_stream));
+  return register_op(ev, device, nullptr, {});
+}
+
+static long long pack_d2h(const std::vector<unsigned long long>& flat, int n,
+                          uintptr_t slab_ptr, uintptr_t pinned_ptr,
+                          unsigned long long total_bytes,
+                          uintptr_t producer_stream, int device) {
+  return launch_pack(flat, n, slab_ptr, pinned_ptr, total_bytes,
+                     producer_stream, device, /*gather=*/true);
+}
+
+static long long scatter_h2d(const std::vector<unsigned long long>& flat, int n,
+                             uintptr_t slab_ptr, uintptr_t pinned_ptr,
+                             unsigned long long total_bytes,
+                             uintptr_t producer_stream, int device) {
+  return launch_pack(flat, n, slab_ptr, pinned_ptr, total_bytes,
+                     producer_stream, device, /*gather=*/false);
+}
+
+static void op_wait(long long handle) {
+  hipEvent_t ev = nullptr;
+  int device = 0;
+  {
+    std::lock_guard<std::mutex> lk(g_mu);
+    auto it = g_ops.find(handle);
+    if (it == g_ops.end()) return;  // already waited
+    ev = it->second.ev;
+    device = it->second.device;
+  }
+  {
+    py::gil_scoped_release release;
+    hipError_t e = hipEventSynchronize(ev);
+    if (e != hipSuccess) {
+      throw std::runtime_error(std::string("hipEventSynchronize: ") +
+                               hipGetErrorString(e));
+    }
+  }
+  std::lock_guard<std::mutex> lk(g_mu);
+  auto it = g_ops.find(handle);
+  if (it != g_ops.end()) {
+    hipEventDestroy(it->second.ev);
+    g_ops.erase(it);
+  }
+}
+
+static bool op_query(long long handle) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  auto it = g_ops.find(handle);
+  if (it == g_ops.end()) return true;
+  return hipEventQuery(it->second.ev) == hipSuccess;
+}
+
+static bool is_managed_ptr(uintptr_t ptr) {
+  hipPointerAttribute_t attr;
+  hipError_t e =
+      hipPointerGetAttributes(&attr, reinterpret_cast<void*>(ptr));
+  if (e != hipSuccess) {
+    (void)hipGetLastError();  // clear
+    return false;
+  }
+  return attr.type == hipMemoryTypeManaged;
+}
+
+static int device_count() {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+PYBIND11_MODULE(_csnap, m) {
+  m.doc() = "torchsnapshot_amd HIP staging engine (gfx950)";
+  m.def("d2h_copy", &d2h_copy, "async D2H copy on the side stream");
+  m.def("h2d_copy", &h2d_copy, "async H2D copy on the side stream");
+  m.def("pack_d2h", &pack_d2h,
+        "gather-pack tensors into a slab and copy it to pinned host memory");
+  m.def("scatter_h2d", &scatter_h2d,
+        "copy pinned host bytes to device and scatter into strided tensors");
+  m.def("wait", &op_wait, "block until an op completes");
+  m.def("query", &op_query, "poll an op");
+  m.def("is_managed_ptr", &is_managed_ptr);
+  m.def("device_count", &device_count);
+}

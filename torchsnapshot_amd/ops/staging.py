@@ -296,6 +296,15 @@ class StagingEngine:
         The caller must ensure the tensors' producing stream is
         torch.cuda.current_stream() of this thread (true for checkpointing:
         tensors are live parameters/opt states, already materialized)."""
+        # the pack kernel indexes within-tensor bytes as u32; non-contiguous
+        # tensors over 2 GiB (pathological: chunking splits along dim 0
+        # upstream) are materialized first
+        tensors = [
+            t
+            if t.is_contiguous() or t.numel() * t.element_size() < 2**31
+            else t.contiguous()
+            for t in tensors
+        ]
         items, offsets, total = build_pack_items(tensors)
         nbytes_list = [it.nbytes for it in items]
         pinned = get_pinned_pool().acquire(max(total, 1))
@@ -326,11 +335,7 @@ class StagingEngine:
         dev_index = self.device.index or 0
         with torch.cuda.device(dev_index):
             cur_stream = torch.cuda.current_stream().cuda_stream
-            single_contig = (
-                len(items) == 1
-                and not items[0].outer_sizes
-                and items[0].vec >= 4
-            )
+            single_contig = len(items) == 1 and not items[0].outer_sizes
             if single_contig:
                 # one SDMA copy, no kernel
                 handle = _csnap.d2h_copy(

@@ -292,7 +292,7 @@ class Snapshot:
                 pg_wrapper,
             )
 
-        write_reqs = _batch(write_reqs)
+        write_reqs = _batch(write_reqs, rank)
 
         global_manifest = cls._gather_manifest(manifest, pg_wrapper)
         metadata = SnapshotMetadata(
@@ -667,10 +667,10 @@ class Snapshot:
         return global_manifest
 
 
-def _batch(write_reqs: List[WriteReq]) -> List[WriteReq]:
+def _batch(write_reqs: List[WriteReq], rank: int = 0) -> List[WriteReq]:
     from .batcher import batch_write_requests
 
-    return batch_write_requests(write_reqs)
+    return batch_write_requests(write_reqs, rank)
 
 
 def _batch_reads(read_reqs: List[ReadReq]) -> List[ReadReq]:
