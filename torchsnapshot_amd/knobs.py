@@ -79,13 +79,34 @@ def get_memory_budget_override_bytes() -> int | None:
 
 
 def get_pinned_block_size_bytes() -> int:
-    """Size of one pinned host staging block in the D2H ring."""
-    return _env_bytes("TSAMD_PINNED_BLOCK_SIZE_BYTES", 256 * _MB)
+    """Size of one pinned host staging block in the D2H ring. Must be at
+    least the max chunk size so chunk stagers hit the pool, not one-off
+    pinned allocations."""
+    return _env_bytes(
+        "TSAMD_PINNED_BLOCK_SIZE_BYTES", max(get_max_chunk_size_bytes(), 512 * _MB)
+    )
+
+
+def get_pinned_pool_bytes() -> int:
+    """Total pinned host memory for the D2H staging ring, per rank. Sized
+    so an async snapshot of a large sharded model stages without stalling:
+    a quarter of host RAM divided among the local ranks, capped at 24 GB."""
+    override = os.environ.get("TSAMD_PINNED_POOL_BYTES")
+    if override is not None:
+        return int(float(override))
+    import psutil
+
+    local_ws = int(os.environ.get("LOCAL_WORLD_SIZE", "1") or "1")
+    total = psutil.virtual_memory().total
+    return min(24 * 1024 * _MB, int(total * 0.25)) // max(local_ws, 1)
 
 
 def get_pinned_block_count() -> int:
     """Number of pinned host staging blocks in the D2H ring."""
-    return _env_int("TSAMD_PINNED_BLOCK_COUNT", 4)
+    override = os.environ.get("TSAMD_PINNED_BLOCK_COUNT")
+    if override is not None:
+        return int(override)
+    return max(get_pinned_pool_bytes() // get_pinned_block_size_bytes(), 2)
 
 
 def is_hip_staging_disabled() -> bool:

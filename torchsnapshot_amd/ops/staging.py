@@ -60,21 +60,21 @@ def _require_ext() -> None:
 
 
 def collapse_layout(t: torch.Tensor) -> Tuple[List[int], List[int]]:
-    """Collapse a tensor's (sizes, strides) by merging dims that are jointly
-    contiguous, dropping size-1 dims. Returns element-unit (sizes, strides),
-    innermost last; a fully contiguous tensor collapses to ([numel], [1])."""
+    """Collapse a tensor's (sizes, strides) by merging ADJACENT LOGICAL dims
+    that are jointly contiguous, dropping size-1 dims. Serialization follows
+    the logical (row-major over t.shape) element order — the same bytes
+    ``t.contiguous()`` would produce — so dims must NOT be reordered by
+    stride. Returns element-unit (sizes, strides), innermost last; a fully
+    contiguous tensor collapses to ([numel], [1])."""
     sizes = [s for s, st in zip(t.shape, t.stride()) if s != 1]
     strides = [st for s, st in zip(t.shape, t.stride()) if s != 1]
     if not sizes:
         # scalar or all-size-1 tensor: one element
         return [1], [1]
-    order = sorted(range(len(sizes)), key=lambda i: (-strides[i], i))
-    sizes = [sizes[i] for i in order]
-    strides = [strides[i] for i in order]
     merged_sizes: List[int] = [sizes[0]]
     merged_strides: List[int] = [strides[0]]
     for s, st in zip(sizes[1:], strides[1:]):
-        # dim fits inside the previous one contiguously?
+        # logical-adjacent dims merge iff outer stride == inner stride * size
         if merged_strides[-1] == st * s:
             merged_sizes[-1] *= s
             merged_strides[-1] = st
