@@ -33,6 +33,16 @@ import torch
 _BASELINE_GBPS = {1: 1.44, 2: 2.29, 4: 3.65, 8: 5.92}
 
 
+def tiny_shapes() -> List[Tuple[str, Tuple[int, ...]]]:
+    """Miniature llama-shaped state for CPU smoke tests of the bench path."""
+    return [
+        ("embed.weight", (64, 16)),
+        ("layer.wq.weight", (16, 16)),
+        ("layer.norm.weight", (16,)),
+        ("head.weight", (64, 16)),
+    ]
+
+
 def llama3_8b_shapes() -> List[Tuple[str, Tuple[int, ...]]]:
     """Parameter shapes of Llama-3-8B (vocab 128256, hidden 4096,
     intermediate 14336, 32 layers, 32 heads / 8 KV heads)."""
@@ -73,9 +83,12 @@ class _BenchState:
 
 
 def build_state(
-    device: torch.device, world_size: int, dtype: torch.dtype
+    device: torch.device,
+    world_size: int,
+    dtype: torch.dtype,
+    model: str = "llama3-8b",
 ) -> Tuple[_BenchState, int]:
-    shapes = llama3_8b_shapes()
+    shapes = tiny_shapes() if model == "tiny" else llama3_8b_shapes()
     total_bytes = 0
     sd: Dict[str, torch.Tensor] = {}
     if world_size > 1:
@@ -83,12 +96,12 @@ def build_state(
         from torch.distributed.tensor import DTensor
         from torch.distributed.tensor.placement_types import Shard
 
-        mesh = init_device_mesh("cuda", (world_size,))
+        mesh = init_device_mesh(device.type, (world_size,))
         for name, shape in shapes:
             assert shape[0] % world_size == 0, (name, shape)
             local_shape = (shape[0] // world_size,) + tuple(shape[1:])
             local = torch.empty(local_shape, dtype=dtype, device=device)
-            local.normal_(0, 0.02)
+            local = local.float().normal_(0, 0.02).to(dtype)
             sd[name] = DTensor.from_local(local, mesh, [Shard(0)])
             total_bytes += int(
                 torch.Size(shape).numel() * local.element_size()
@@ -113,6 +126,10 @@ def main() -> None:
     parser.add_argument(
         "--keep", action="store_true", help="keep the checkpoint directory"
     )
+    parser.add_argument("--device", choices=["cuda", "cpu"], default="cuda")
+    parser.add_argument(
+        "--model", choices=["llama3-8b", "tiny"], default="llama3-8b"
+    )
     args = parser.parse_args()
 
     from torchsnapshot_amd import Snapshot
@@ -120,13 +137,18 @@ def main() -> None:
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = args.device == "cuda"
     if world_size > 1:
         import torch.distributed as dist
 
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl")
-    device = torch.device("cuda", local_rank)
-    torch.cuda.set_device(device)
+        if use_cuda:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+    if use_cuda:
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
 
     bench_dir = args.dir or "/tmp/tsamd_bench"
     ckpt_path = os.path.join(bench_dir, "ckpt")
@@ -135,7 +157,9 @@ def main() -> None:
         shutil.rmtree(ckpt_path, ignore_errors=True)
     _barrier(world_size)
 
-    state, total_bytes = build_state(device, world_size, torch.bfloat16)
+    state, total_bytes = build_state(
+        device, world_size, torch.bfloat16, model=args.model
+    )
     app_state = {"model": state}
 
     # warmup (untimed): allocates pinned blocks, compiles nothing, warms
@@ -143,12 +167,14 @@ def main() -> None:
     for _ in range(args.warmup):
         Snapshot.take(ckpt_path, app_state)
     _barrier(world_size)
-    torch.cuda.synchronize(device)
+    if use_cuda:
+        torch.cuda.synchronize(device)
 
     t0 = time.monotonic()
     for _ in range(args.steps):
         Snapshot.take(ckpt_path, app_state)
-    torch.cuda.synchronize(device)
+    if use_cuda:
+        torch.cuda.synchronize(device)
     _barrier(world_size)
     elapsed = time.monotonic() - t0
     elapsed = _max_over_ranks(elapsed, world_size, device)
@@ -184,7 +210,7 @@ def main() -> None:
             "data": "synthetic",
             "stall_sec": round(stall_s, 3),
             "config": {
-                "model": "llama-3-8b",
+                "model": "llama-3-8b" if args.model != "tiny" else "tiny",
                 "model_bytes": total_bytes,
                 "global_batch": None,
                 "seq_len": None,
@@ -212,7 +238,8 @@ def _max_over_ranks(value: float, world_size: int, device) -> float:
         return value
     import torch.distributed as dist
 
-    t = torch.tensor([value], dtype=torch.float64, device=device)
+    dev = device if device.type == "cuda" else torch.device("cpu")
+    t = torch.tensor([value], dtype=torch.float64, device=dev)
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     return float(t.item())
 
