@@ -137,6 +137,19 @@ class ShardConsumer(BufferConsumer):
         def work() -> None:
             if self.shard_entry.serializer == "torch_save":
                 shard = torch_load_from_bytes(bytes(buf))
+            elif all(
+                dst.device.type == "cuda" for dst, _ in self.targets
+            ) and self.targets:
+                # device restore: move the whole persisted shard up once
+                # (pinned bounce + SDMA), scatter overlaps on the GPU
+                from ..ops.staging import copy_buffer_via_pinned
+
+                shard = copy_buffer_via_pinned(
+                    buf,
+                    dtype=str_to_dtype(self.shard_entry.dtype),
+                    shape=tuple(self.shard_entry.shape),
+                    device=self.targets[0][0].device,
+                )
             else:
                 shard = tensor_from_memoryview(
                     memoryview(buf),

@@ -300,6 +300,20 @@ class TensorBufferConsumer(BufferConsumer):
         def work() -> None:
             if self.entry.serializer == SERIALIZER_TORCH_SAVE:
                 loaded = torch_load_from_bytes(bytes(buf))
+            elif (
+                self.tensor_out is not None
+                and self.tensor_out.device.type == "cuda"
+            ):
+                # device restore fast path: pinned bounce + SDMA H2D, then
+                # any strided/cast scatter runs ON the GPU
+                from ..ops.staging import copy_buffer_via_pinned
+
+                loaded = copy_buffer_via_pinned(
+                    buf,
+                    dtype=str_to_dtype(self.entry.dtype),
+                    shape=tuple(self.entry.shape),
+                    device=self.tensor_out.device,
+                )
             else:
                 dtype = str_to_dtype(self.entry.dtype)
                 loaded = tensor_from_memoryview(

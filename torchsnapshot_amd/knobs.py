@@ -98,7 +98,7 @@ def get_pinned_pool_bytes() -> int:
 
     local_ws = int(os.environ.get("LOCAL_WORLD_SIZE", "1") or "1")
     total = psutil.virtual_memory().total
-    return min(24 * 1024 * _MB, int(total * 0.25)) // max(local_ws, 1)
+    return min(24 * 1024 * _MB, int(total * 0.35)) // max(local_ws, 1)
 
 
 def get_pinned_block_count() -> int:

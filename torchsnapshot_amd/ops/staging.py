@@ -28,6 +28,7 @@ import threading
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence, Tuple
 
+import numpy as np
 import torch
 
 from .. import knobs
@@ -385,12 +386,41 @@ class StagingEngine:
 
 
 def _pack_mode() -> str:
+    # "direct": the gather kernel writes straight into pinned host memory
+    # over PCIe (measured ~50 GB/s on MI355X, and +20% on the full
+    # checkpoint bench vs the device-slab + SDMA hop, which also costs a
+    # transient slab allocation per batch). "slab" keeps CU time minimal
+    # (kernel runs at HBM speed, SDMA does the PCIe leg) for overlap with
+    # heavy training compute.
     import os
 
-    mode = os.environ.get("TSAMD_STAGE_MODE", "slab")
+    mode = os.environ.get("TSAMD_STAGE_MODE", "direct")
     if mode not in ("slab", "direct"):
         raise ValueError(f"TSAMD_STAGE_MODE must be slab|direct, got {mode}")
     return mode
+
+
+def copy_buffer_via_pinned(
+    buf, dtype: torch.dtype, shape: Sequence[int], device: torch.device
+) -> torch.Tensor:
+    """Restore fast path: host buffer -> pinned bounce -> SDMA H2D.
+
+    Returns a contiguous device tensor with the buffer's content. The
+    pageable-copy alternative (torch.frombuffer(...).to(cuda)) measures
+    ~3x slower on MI355X."""
+    mv = memoryview(buf)
+    nbytes = mv.nbytes
+    pool = get_pinned_pool()
+    block = pool.acquire(max(nbytes, 1))
+    try:
+        pinned_np = block.tensor.numpy()
+        pinned_np[:nbytes] = np.frombuffer(mv, dtype=np.uint8)
+        dev_u8 = block.tensor[:nbytes].to(device, non_blocking=False)
+    finally:
+        pool.release(block)
+    if dtype == torch.uint8:
+        return dev_u8.reshape(tuple(shape))
+    return dev_u8.view(dtype).reshape(tuple(shape))
 
 
 _engines: Dict[int, StagingEngine] = {}
