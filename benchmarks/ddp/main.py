@@ -70,6 +70,9 @@ def main() -> None:
         )
 
     if args.compare_torch_save and rank == 0:
+        # free the snapshot's disk space first: the torch.save copy of a
+        # 20 GB model must not race the snapshot for a small local disk
+        shutil.rmtree(path, ignore_errors=True)
         t0 = time.monotonic()
         torch.save(sd.state_dict(), os.path.join(args.work_dir, "torch_save.pt"))
         elapsed = time.monotonic() - t0

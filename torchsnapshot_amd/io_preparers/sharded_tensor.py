@@ -129,6 +129,27 @@ class ShardConsumer(BufferConsumer):
     ) -> None:
         self.shard_entry = shard_entry
         self.targets = targets
+        self._pinned_block = None
+        self._pinned_nbytes = 0
+
+    def all_targets_on_device(self) -> bool:
+        return bool(self.targets) and all(
+            dst.device.type == "cuda" for dst, _ in self.targets
+        )
+
+    def alloc_pinned_buffer(self, nbytes: int) -> memoryview:
+        from ..ops.staging import get_pinned_pool
+
+        self._pinned_block = get_pinned_pool().acquire(max(nbytes, 1))
+        self._pinned_nbytes = nbytes
+        return memoryview(self._pinned_block.tensor.numpy())[:nbytes]
+
+    def close(self) -> None:
+        if self._pinned_block is not None:
+            from ..ops.staging import get_pinned_pool
+
+            get_pinned_pool().release(self._pinned_block)
+            self._pinned_block = None
 
     def get_consuming_cost_bytes(self) -> int:
         return self.shard_entry.nbytes_estimate()
@@ -137,11 +158,21 @@ class ShardConsumer(BufferConsumer):
         def work() -> None:
             if self.shard_entry.serializer == "torch_save":
                 shard = torch_load_from_bytes(bytes(buf))
-            elif all(
-                dst.device.type == "cuda" for dst, _ in self.targets
-            ) and self.targets:
-                # device restore: move the whole persisted shard up once
-                # (pinned bounce + SDMA), scatter overlaps on the GPU
+            elif self._pinned_block is not None and self.all_targets_on_device():
+                # storage read landed in pinned memory: straight SDMA H2D,
+                # overlap scatter runs on the GPU
+                n = self._pinned_nbytes
+                dtype = str_to_dtype(self.shard_entry.dtype)
+                dev_u8 = self._pinned_block.tensor[:n].to(
+                    self.targets[0][0].device, non_blocking=False
+                )
+                shard = (
+                    dev_u8.view(dtype).reshape(tuple(self.shard_entry.shape))
+                    if dtype != torch.uint8
+                    else dev_u8.reshape(tuple(self.shard_entry.shape))
+                )
+            elif self.all_targets_on_device():
+                # byte-range/batched path: pinned bounce + SDMA
                 from ..ops.staging import copy_buffer_via_pinned
 
                 shard = copy_buffer_via_pinned(
@@ -186,11 +217,21 @@ def plan_shard_reads(
         byte_range = (
             tuple(shard.tensor.byte_range) if shard.tensor.byte_range else None
         )
+        consumer = ShardConsumer(shard_entry=shard.tensor, targets=hits)
+        buf_alloc = (
+            consumer.alloc_pinned_buffer
+            if (
+                consumer.all_targets_on_device()
+                and shard.tensor.serializer != "torch_save"
+            )
+            else None
+        )
         read_reqs.append(
             ReadReq(
                 path=shard.tensor.location,
-                consumer=ShardConsumer(shard_entry=shard.tensor, targets=hits),
+                consumer=consumer,
                 byte_range=byte_range,
+                buf_alloc=buf_alloc,
             )
         )
     return read_reqs

@@ -116,8 +116,25 @@ class TensorIOPreparer:
         fut = LoadFuture(tensor_out)
         consumer = TensorBufferConsumer(entry=entry, tensor_out=tensor_out, fut=fut)
         byte_range = tuple(entry.byte_range) if entry.byte_range else None
+        # device targets read straight into pinned memory (if the request
+        # ends up merged into a batched span, the span read ignores this
+        # and the consumer falls back to the bounce path)
+        buf_alloc = (
+            consumer.alloc_pinned_buffer
+            if (
+                tensor_out is not None
+                and tensor_out.device.type == "cuda"
+                and entry.serializer == SERIALIZER_BUFFER
+            )
+            else None
+        )
         return [
-            ReadReq(path=entry.location, consumer=consumer, byte_range=byte_range)
+            ReadReq(
+                path=entry.location,
+                consumer=consumer,
+                byte_range=byte_range,
+                buf_alloc=buf_alloc,
+            )
         ], fut
 
     @staticmethod
@@ -289,6 +306,24 @@ class TensorBufferConsumer(BufferConsumer):
         self.entry = entry
         self.tensor_out = tensor_out
         self.fut = fut
+        self._pinned_block = None
+        self._pinned_nbytes = 0
+
+    def alloc_pinned_buffer(self, nbytes: int) -> memoryview:
+        """ReadReq.buf_alloc hook: the storage layer reads directly into
+        pinned memory, so H2D needs no bounce copy."""
+        from ..ops.staging import get_pinned_pool
+
+        self._pinned_block = get_pinned_pool().acquire(max(nbytes, 1))
+        self._pinned_nbytes = nbytes
+        return memoryview(self._pinned_block.tensor.numpy())[:nbytes]
+
+    def close(self) -> None:
+        if self._pinned_block is not None:
+            from ..ops.staging import get_pinned_pool
+
+            get_pinned_pool().release(self._pinned_block)
+            self._pinned_block = None
 
     def get_consuming_cost_bytes(self) -> int:
         nbytes = self.entry.nbytes_estimate()
@@ -298,24 +333,38 @@ class TensorBufferConsumer(BufferConsumer):
 
     async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
         def work() -> None:
+            dtype = str_to_dtype(self.entry.dtype) if not self.entry.serializer == SERIALIZER_TORCH_SAVE else None
             if self.entry.serializer == SERIALIZER_TORCH_SAVE:
                 loaded = torch_load_from_bytes(bytes(buf))
+            elif (
+                self._pinned_block is not None
+                and self.tensor_out is not None
+                and self.tensor_out.device.type == "cuda"
+            ):
+                # buffer already lives in pinned memory: straight SDMA H2D
+                n = self._pinned_nbytes
+                dev_u8 = self._pinned_block.tensor[:n].to(
+                    self.tensor_out.device, non_blocking=False
+                )
+                loaded = (
+                    dev_u8.view(dtype).reshape(tuple(self.entry.shape))
+                    if dtype != torch.uint8
+                    else dev_u8.reshape(tuple(self.entry.shape))
+                )
             elif (
                 self.tensor_out is not None
                 and self.tensor_out.device.type == "cuda"
             ):
-                # device restore fast path: pinned bounce + SDMA H2D, then
-                # any strided/cast scatter runs ON the GPU
+                # batched-span slice: pinned bounce + SDMA H2D
                 from ..ops.staging import copy_buffer_via_pinned
 
                 loaded = copy_buffer_via_pinned(
                     buf,
-                    dtype=str_to_dtype(self.entry.dtype),
+                    dtype=dtype,
                     shape=tuple(self.entry.shape),
                     device=self.tensor_out.device,
                 )
             else:
-                dtype = str_to_dtype(self.entry.dtype)
                 loaded = tensor_from_memoryview(
                     memoryview(buf), dtype=dtype, shape=tuple(self.entry.shape)
                 )

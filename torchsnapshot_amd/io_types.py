@@ -53,6 +53,10 @@ class BufferConsumer(abc.ABC):
     def get_consuming_cost_bytes(self) -> int:
         """Peak host-memory cost of holding + consuming this buffer."""
 
+    def close(self) -> None:
+        """Called when the pipeline is done with this request (success or
+        failure); return pooled buffers here."""
+
 
 @dataclass
 class WriteReq:
@@ -68,6 +72,10 @@ class ReadReq:
     path: str
     consumer: BufferConsumer
     byte_range: Optional[Tuple[int, int]] = None
+    # Optional allocator for the read buffer (nbytes -> writable
+    # memoryview). Device-targeted consumers hand out pinned memory here so
+    # storage reads land directly in DMA-able pages (no bounce copy).
+    buf_alloc: Optional[Any] = None
 
 
 @dataclass
@@ -81,6 +89,7 @@ class ReadIO:
     path: str
     byte_range: Optional[Tuple[int, int]] = None
     buf: Optional[BufferType] = field(default=None)
+    buf_alloc: Optional[Any] = None
 
 
 class StoragePlugin(abc.ABC):

@@ -413,8 +413,8 @@ def copy_buffer_via_pinned(
     pool = get_pinned_pool()
     block = pool.acquire(max(nbytes, 1))
     try:
-        pinned_np = block.tensor.numpy()
-        pinned_np[:nbytes] = np.frombuffer(mv, dtype=np.uint8)
+        src = torch.frombuffer(mv, dtype=torch.uint8)
+        block.tensor[:nbytes].copy_(src)  # releases the GIL
         dev_u8 = block.tensor[:nbytes].to(device, non_blocking=False)
     finally:
         pool.release(block)

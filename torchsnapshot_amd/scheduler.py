@@ -313,7 +313,11 @@ def execute_read_reqs(
             cost = req.consumer.get_consuming_cost_bytes()
             await budget.acquire(cost)
             try:
-                read_io = ReadIO(path=req.path, byte_range=req.byte_range)
+                read_io = ReadIO(
+                    path=req.path,
+                    byte_range=req.byte_range,
+                    buf_alloc=req.buf_alloc,
+                )
                 async with io_sem:
                     await storage.read(read_io)
                 buf = read_io.buf
@@ -321,6 +325,7 @@ def execute_read_reqs(
                 await req.consumer.consume_buffer(ctx, buf)
                 stats.done_reqs += 1
             finally:
+                req.consumer.close()
                 await budget.release(cost)
 
         tasks = [asyncio.create_task(handle(r)) for r in ordered]

@@ -65,8 +65,13 @@ class FSStoragePlugin(StoragePlugin):
         else:
             start, end = read_io.byte_range
         nbytes = end - start
-        buf = bytearray(nbytes)
+        if read_io.buf_alloc is not None:
+            buf = read_io.buf_alloc(nbytes)
+        else:
+            buf = bytearray(nbytes)
         mv = memoryview(buf)
+        if mv.format != "B":
+            mv = mv.cast("B")
         fd = os.open(full, os.O_RDONLY)
         try:
             off = 0
