@@ -148,11 +148,11 @@ class TensorIOPreparer:
         dtype = str_to_dtype(entry.dtype)
         nbytes = entry.nbytes_estimate()
         elem = max(dtype.itemsize, 1)
-        # staging target: load flat, copy into tensor_out at the end when the
-        # output isn't a directly-addressable contiguous CPU tensor
+        # tiles write straight into the output when it is contiguous with a
+        # matching dtype — on ANY device (CUDA tiles go pinned -> SDMA H2D,
+        # so peak host memory stays at the budget, never the tensor size)
         direct = (
             tensor_out is not None
-            and tensor_out.device.type == "cpu"
             and tensor_out.is_contiguous()
             and tensor_out.dtype == dtype
         )
@@ -172,12 +172,18 @@ class TensorIOPreparer:
         )
         for start in range(0, nbytes, tile):
             end = min(start + tile, nbytes)
+            consumer = _TensorTileConsumer(
+                dst_flat_u8=flat_u8, start=start, end=end, state=state
+            )
             reqs.append(
                 ReadReq(
                     path=entry.location,
                     byte_range=(start, end),
-                    consumer=_TensorTileConsumer(
-                        dst_flat_u8=flat_u8, start=start, end=end, state=state
+                    consumer=consumer,
+                    buf_alloc=(
+                        consumer.alloc_pinned_buffer
+                        if staging.device.type == "cuda"
+                        else None
                     ),
                 )
             )
@@ -212,13 +218,31 @@ class _TensorTileConsumer(BufferConsumer):
         self.start = start
         self.end = end
         self.state = state
+        self._pinned_block = None
+
+    def alloc_pinned_buffer(self, nbytes: int) -> memoryview:
+        from ..ops.staging import get_pinned_pool
+
+        self._pinned_block = get_pinned_pool().acquire(max(nbytes, 1))
+        return memoryview(self._pinned_block.tensor.numpy())[:nbytes]
+
+    def close(self) -> None:
+        if self._pinned_block is not None:
+            from ..ops.staging import get_pinned_pool
+
+            get_pinned_pool().release(self._pinned_block)
+            self._pinned_block = None
 
     def get_consuming_cost_bytes(self) -> int:
         return self.end - self.start
 
     async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
         def work() -> None:
-            src = torch.frombuffer(buf, dtype=torch.uint8)
+            n = self.end - self.start
+            if self._pinned_block is not None:
+                src = self._pinned_block.tensor[:n]  # pinned: async-capable
+            else:
+                src = torch.frombuffer(buf, dtype=torch.uint8)
             self.dst_flat_u8[self.start : self.end].copy_(src)
             with self.state.lock:
                 self.state.remaining -= 1

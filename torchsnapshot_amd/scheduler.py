@@ -77,6 +77,11 @@ class ExecutionStats:
     done_reqs: int = 0
     staged_bytes: int = 0
     io_bytes: int = 0
+    # cumulative per-phase busy time across requests (can exceed wall
+    # time with concurrency); enable logging with TSAMD_TIMING=1
+    stage_s: float = 0.0
+    io_s: float = 0.0
+    consume_s: float = 0.0
     begin_ts: float = field(default_factory=time.monotonic)
     staged_ts: Optional[float] = None
     end_ts: Optional[float] = None
@@ -222,7 +227,9 @@ def execute_write_reqs(
             await budget.acquire(cost)
             try:
                 async with staging_sem:
+                    t0 = time.monotonic()
                     buf = await req.stager.stage_buffer(ctx)
+                    stats.stage_s += time.monotonic() - t0
                 nbytes = memoryview(buf).nbytes
                 stats.staged_reqs += 1
                 stats.staged_bytes += nbytes
@@ -230,7 +237,9 @@ def execute_write_reqs(
                 if staged_remaining == 0:
                     all_staged.set()
                 async with io_sem:
+                    t0 = time.monotonic()
                     await storage.write(WriteIO(path=req.path, buf=buf))
+                    stats.io_s += time.monotonic() - t0
                 stats.io_bytes += nbytes
                 stats.done_reqs += 1
                 req.stager.release_buffer()
@@ -269,6 +278,16 @@ def sync_execute_write_reqs(
         write_reqs, storage, memory_budget_bytes, rank=rank, is_async=False
     )
     pending.complete()
+    import os
+
+    if os.environ.get("TSAMD_TIMING"):
+        s = pending.stats
+        wall = (s.end_ts or 0) - s.begin_ts
+        print(
+            f"[tsamd timing] write: wall={wall:.2f}s reqs={s.total_reqs} "
+            f"bytes={s.io_bytes/1e9:.2f}GB stage_busy={s.stage_s:.2f}s "
+            f"io_busy={s.io_s:.2f}s"
+        )
     if rank == 0 and pending.stats.io_bytes:
         logger.info(
             "Wrote %.1f MB in %.2fs (%.2f GB/s)",
@@ -319,10 +338,14 @@ def execute_read_reqs(
                     buf_alloc=req.buf_alloc,
                 )
                 async with io_sem:
+                    t0 = time.monotonic()
                     await storage.read(read_io)
+                    stats.io_s += time.monotonic() - t0
                 buf = read_io.buf
                 stats.io_bytes += memoryview(buf).nbytes
+                t0 = time.monotonic()
                 await req.consumer.consume_buffer(ctx, buf)
+                stats.consume_s += time.monotonic() - t0
                 stats.done_reqs += 1
             finally:
                 req.consumer.close()
@@ -350,6 +373,16 @@ def sync_execute_read_reqs(
 ) -> ExecutionStats:
     pending = execute_read_reqs(read_reqs, storage, memory_budget_bytes, rank)
     pending.complete()
+    import os
+
+    if os.environ.get("TSAMD_TIMING"):
+        s = pending.stats
+        wall = (s.end_ts or 0) - s.begin_ts
+        print(
+            f"[tsamd timing] read: wall={wall:.2f}s reqs={s.total_reqs} "
+            f"bytes={s.io_bytes/1e9:.2f}GB io_busy={s.io_s:.2f}s "
+            f"consume_busy={s.consume_s:.2f}s"
+        )
     if rank == 0 and pending.stats.io_bytes:
         logger.info(
             "Read %.1f MB in %.2fs (%.2f GB/s)",
