@@ -156,6 +156,36 @@ def _merge_sharded_entries(entries: List[Entry]) -> Entry:
     raise TypeError(f"cannot merge entries of type {type(first).__name__}")
 
 
+def remove_entry_from_manifest(manifest: Manifest, path: str) -> None:
+    """Remove an entry AND unlink it from its parent container entry so the
+    manifest stays inflatable (parity with reference
+    torchsnapshot/manifest_ops.py:250-288)."""
+    from .flatten import unescape_key
+
+    manifest.pop(path, None)
+    if "/" not in path:
+        return
+    parent_path, _, segment = path.rpartition("/")
+    parent = manifest.get(parent_path)
+    if parent is None:
+        return
+    from .manifest import DictEntry, ListEntry
+
+    if isinstance(parent, DictEntry):  # covers OrderedDictEntry
+        key = unescape_key(segment)
+        for cand in (key, int(key) if key.lstrip("-").isdigit() else key):
+            if cand in parent.keys:
+                parent.keys.remove(cand)
+                break
+    elif isinstance(parent, ListEntry):
+        # a list with a removed element cannot be reindexed faithfully;
+        # degrade the parent to a dict-like view is not possible either —
+        # refuse, matching the reference's container constraints
+        raise ValueError(
+            f"cannot remove '{path}': parent '{parent_path}' is a list"
+        )
+
+
 def handle_sharded_tensor_elasticity(
     rank_manifest: Manifest,
     payload_entries: Dict[str, Entry],
@@ -190,7 +220,7 @@ def handle_sharded_tensor_elasticity(
                 path,
             )
             del payload_entries[path]
-            rank_manifest.pop(path, None)
+            remove_entry_from_manifest(rank_manifest, path)
 
     for path, obj in list(flattened_target.items()):
         if _runtime_is_sharded(obj) and path not in payload_entries:

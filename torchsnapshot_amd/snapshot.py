@@ -437,13 +437,17 @@ class Snapshot:
         _, flattened_tgt = flatten(sd, prefix=key)
 
         prefix = f"{key}/"
+        # deep-copied: elasticity handling may prune entries/container keys,
+        # and the cached metadata must stay pristine
+        import copy as _copy
+
         selected = {
             p: e
             for p, e in payload_entries.items()
             if p == key or p.startswith(prefix)
         }
         sub_manifest = {
-            p: e
+            p: _copy.deepcopy(e)
             for p, e in rank_manifest.items()
             if p == key or p.startswith(prefix)
         }
