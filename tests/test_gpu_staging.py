@@ -172,3 +172,27 @@ def test_async_take_gpu():
         out = StateDict(w=torch.zeros(512, 512, device="cuda"))
         snap.restore({"sd": out})
         assert torch.equal(out["w"], saved)
+
+
+def test_uvm_managed_tensor():
+    """hipMallocManaged tensor: CPU-addressable, staged zero-copy."""
+    from torchsnapshot_amd.uvm_tensor import (
+        is_uvm_tensor,
+        new_managed_tensor,
+        prefetch_to_device,
+        uvm_to_cpu,
+    )
+
+    t = new_managed_tensor((128, 64), dtype=torch.float32)
+    t.uniform_(-1, 1)
+    # the CPU view addresses managed pages
+    assert t.device.type == "cpu"
+    prefetch_to_device(t, 0)
+    prefetch_to_device(t, -1)
+    # snapshot it and restore
+    sd = StateDict(emb=t)
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict(emb=torch.zeros(128, 64))
+        snap.restore({"sd": out})
+        assert torch.equal(out["emb"], t)

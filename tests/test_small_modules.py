@@ -1,0 +1,135 @@
+"""Coverage for the small cross-cutting modules: memoryview stream,
+events, glob matching, knobs overrides, rss profiler, pg_wrapper no-dist
+behavior."""
+
+import io
+import time
+from collections import deque
+
+import pytest
+import torch
+
+from torchsnapshot_amd import knobs
+from torchsnapshot_amd.event import (
+    Event,
+    log_event,
+    register_event_handler,
+    unregister_event_handler,
+)
+from torchsnapshot_amd.memoryview_stream import MemoryviewStream
+from torchsnapshot_amd.pg_wrapper import PGWrapper
+from torchsnapshot_amd.rss_profiler import max_rss_delta_mb, measure_rss_deltas
+from torchsnapshot_amd.snapshot import glob_match
+
+
+def test_memoryview_stream():
+    data = bytes(range(256))
+    s = MemoryviewStream(memoryview(data))
+    assert s.readable() and s.seekable()
+    assert s.read(10) == data[:10]
+    assert s.tell() == 10
+    s.seek(100)
+    assert s.read(5) == data[100:105]
+    s.seek(-6, io.SEEK_END)
+    assert s.read() == data[-6:]
+    s.seek(0)
+    buf = bytearray(300)
+    assert s.readinto(buf) == 256
+    assert bytes(buf[:256]) == data
+    s.close()
+    with pytest.raises(ValueError):
+        s.read(1)
+
+
+def test_glob_match():
+    assert glob_match("model/lin.weight", "model/**")
+    assert glob_match("model/a/b/c", "**")
+    assert glob_match("model/lin.weight", "model/*.weight")
+    assert not glob_match("model/sub/lin.weight", "model/*.weight")
+    assert glob_match("model/sub/lin.weight", "model/**/*.weight")
+    assert glob_match("x", "**")
+    assert not glob_match("model/x", "optim/**")
+    assert glob_match("a/b", "a/b")
+
+
+def test_event_handlers():
+    events = []
+
+    class Handler:
+        def handle_event(self, event: Event) -> None:
+            events.append(event)
+
+    h = Handler()
+    register_event_handler(h)
+    try:
+        log_event(Event("test_event", {"k": 1}))
+    finally:
+        unregister_event_handler(h)
+    assert events and events[0].name == "test_event"
+    log_event(Event("after", {}))
+    assert len(events) == 1  # unregistered
+
+
+def test_events_emitted_by_take(tmp_path):
+    from torchsnapshot_amd import Snapshot, StateDict
+
+    events = []
+
+    class Handler:
+        def handle_event(self, event: Event) -> None:
+            events.append(event.name)
+
+    h = Handler()
+    register_event_handler(h)
+    try:
+        snap = Snapshot.take(str(tmp_path / "s"), {"sd": StateDict(a=1)})
+        snap.restore({"sd": StateDict()})
+    finally:
+        unregister_event_handler(h)
+    assert "take_start" in events and "take_end" in events
+    assert "restore_start" in events and "restore_end" in events
+
+
+def test_knobs_overrides():
+    with knobs.override_max_chunk_size_bytes(123):
+        assert knobs.get_max_chunk_size_bytes() == 123
+    assert knobs.get_max_chunk_size_bytes() != 123
+    with knobs.override_batching_disabled(True):
+        assert knobs.is_batching_disabled()
+    assert not knobs.is_batching_disabled()
+    with knobs.override_max_io_concurrency(3):
+        assert knobs.get_max_io_concurrency() == 3
+
+
+def test_rss_profiler():
+    deltas = deque()
+    with measure_rss_deltas(deltas, interval_s=0.01):
+        blob = bytearray(64 * 1024 * 1024)
+        blob[::4096] = b"x" * len(blob[::4096])
+        time.sleep(0.1)
+    assert len(deltas) > 0
+    assert max_rss_delta_mb(deltas) >= 0
+
+
+def test_pg_wrapper_no_dist():
+    pgw = PGWrapper(None)
+    assert pgw.get_rank() == 0
+    assert pgw.get_world_size() == 1
+    pgw.barrier()
+    out = [None]
+    pgw.all_gather_object(out, "x")
+    assert out == ["x"]
+    lst = ["payload"]
+    pgw.broadcast_object_list(lst)
+    assert lst == ["payload"]
+    res = [None]
+    pgw.scatter_object_list(res, ["only"])
+    assert res == ["only"]
+
+
+def test_uvm_fallbacks_on_cpu():
+    from torchsnapshot_amd.uvm_tensor import is_uvm_tensor, uvm_to_cpu
+
+    t = torch.rand(4)
+    assert not is_uvm_tensor(t)
+    assert uvm_to_cpu(t) is t

@@ -280,12 +280,20 @@ class TensorBufferStager(BufferStager):
 
     def _stage_device(self) -> BufferType:
         from ..ops.staging import get_staging_engine
+        from ..uvm_tensor import is_uvm_tensor, uvm_to_cpu
 
         t = self.tensor.detach()
         if t.is_quantized or self.serializer == SERIALIZER_TORCH_SAVE:
             # rare path: bring to host with torch, then pickle
             cpu = t.cpu()
             return torch_save_as_bytes(cpu)
+        if is_uvm_tensor(t):
+            # managed memory is CPU-addressable: serialize zero-copy, no
+            # D2H needed (clone when async: training may mutate the pages)
+            cpu = uvm_to_cpu(t)
+            if self.is_async_snapshot:
+                cpu = cpu.clone()
+            return tensor_as_memoryview(cpu.contiguous())
         engine = get_staging_engine(t.device)
         batch = engine.stage([t])
         batch.wait()  # blocks in executor thread; GIL released inside HIP

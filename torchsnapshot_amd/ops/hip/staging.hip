@@ -455,6 +455,39 @@ static int device_count() {
   return n;
 }
 
+// -- UVM (managed memory) ----------------------------------------------------
+
+static uintptr_t managed_alloc(unsigned long long nbytes, int device) {
+  HIP_CHECK(hipSetDevice(device));
+  void* ptr = nullptr;
+  HIP_CHECK(hipMallocManaged(&ptr, nbytes, hipMemAttachGlobal));
+  return reinterpret_cast<uintptr_t>(ptr);
+}
+
+static void managed_free(uintptr_t ptr) {
+  HIP_CHECK(hipFree(reinterpret_cast<void*>(ptr)));
+}
+
+static void managed_advise_preferred_cpu(uintptr_t ptr,
+                                         unsigned long long nbytes) {
+  // embedding-table pattern: keep pages host-resident, GPU reads over PCIe
+  hipError_t e = hipMemAdvise(reinterpret_cast<void*>(ptr), nbytes,
+                              hipMemAdviseSetPreferredLocation,
+                              hipCpuDeviceId);
+  (void)e;  // advisory only
+}
+
+static void managed_prefetch(uintptr_t ptr, unsigned long long nbytes,
+                             int device) {
+  // device = -1 prefetches to the CPU
+  DeviceCtx& ctx = get_ctx(device >= 0 ? device : 0);
+  hipError_t e = hipMemPrefetchAsync(
+      reinterpret_cast<void*>(ptr), nbytes,
+      device >= 0 ? device : hipCpuDeviceId, ctx.h2d_stream);
+  (void)e;  // advisory only
+  (void)hipStreamSynchronize(ctx.h2d_stream);
+}
+
 PYBIND11_MODULE(_csnap, m) {
   m.doc() = "torchsnapshot_amd HIP staging engine (gfx950)";
   m.def("d2h_copy", &d2h_copy, "async D2H copy on the side stream");
@@ -467,4 +500,8 @@ PYBIND11_MODULE(_csnap, m) {
   m.def("query", &op_query, "poll an op");
   m.def("is_managed_ptr", &is_managed_ptr);
   m.def("device_count", &device_count);
+  m.def("managed_alloc", &managed_alloc, "hipMallocManaged");
+  m.def("managed_free", &managed_free);
+  m.def("managed_advise_preferred_cpu", &managed_advise_preferred_cpu);
+  m.def("managed_prefetch", &managed_prefetch);
 }
