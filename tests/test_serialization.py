@@ -49,8 +49,10 @@ def test_dtype_string_round_trip():
 
 
 def test_pick_serializer():
+    from torchsnapshot_amd.serialization import SERIALIZER_QTENSOR
+
     assert pick_serializer(torch.rand(3)) == SERIALIZER_BUFFER
-    assert pick_serializer(rand_tensor((3,), torch.qint8)) == SERIALIZER_TORCH_SAVE
+    assert pick_serializer(rand_tensor((3,), torch.qint8)) == SERIALIZER_QTENSOR
 
 
 def test_scalar_tensor():
@@ -95,3 +97,32 @@ def test_memoryview_zero_copy():
     # zero-copy: mutation visible through the view
     t2 = tensor_from_memoryview(mv, torch.float32, (128,))
     assert t2[0].item() == 42.0
+
+
+def test_qtensor_layout_round_trip():
+    from torchsnapshot_amd.serialization import (
+        qtensor_as_bytes,
+        qtensor_from_bytes,
+    )
+
+    for dtype in (torch.qint8, torch.quint8, torch.qint32):
+        t = torch.quantize_per_tensor(
+            torch.rand(13, 7), scale=0.07, zero_point=2, dtype=dtype
+        )
+        t2 = qtensor_from_bytes(qtensor_as_bytes(t))
+        assert tensor_eq(t, t2)
+        assert t2.q_scale() == t.q_scale()
+        assert t2.q_zero_point() == t.q_zero_point()
+
+    # per-channel
+    t = torch.quantize_per_channel(
+        torch.rand(6, 10),
+        scales=torch.rand(6) * 0.1 + 0.01,
+        zero_points=torch.randint(0, 10, (6,)),
+        axis=0,
+        dtype=torch.qint8,
+    )
+    t2 = qtensor_from_bytes(qtensor_as_bytes(t))
+    assert tensor_eq(t, t2)
+    assert torch.equal(t2.q_per_channel_scales(), t.q_per_channel_scales())
+    assert t2.q_per_channel_axis() == 0

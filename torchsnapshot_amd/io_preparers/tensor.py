@@ -29,9 +29,12 @@ from ..io_types import (
 from ..manifest import TensorEntry
 from ..serialization import (
     SERIALIZER_BUFFER,
+    SERIALIZER_QTENSOR,
     SERIALIZER_TORCH_SAVE,
     dtype_to_str,
     pick_serializer,
+    qtensor_as_bytes,
+    qtensor_from_bytes,
     str_to_dtype,
     tensor_as_memoryview,
     tensor_from_memoryview,
@@ -267,8 +270,8 @@ class TensorBufferStager(BufferStager):
 
     def get_staging_cost_bytes(self) -> int:
         nbytes = self.tensor.numel() * self.tensor.element_size()
-        if self.serializer == SERIALIZER_TORCH_SAVE:
-            # torch.save materializes a second copy while pickling
+        if self.serializer in (SERIALIZER_TORCH_SAVE, SERIALIZER_QTENSOR):
+            # these serializers materialize a second copy while packing
             return 2 * nbytes
         return nbytes
 
@@ -283,9 +286,11 @@ class TensorBufferStager(BufferStager):
         from ..uvm_tensor import is_uvm_tensor, uvm_to_cpu
 
         t = self.tensor.detach()
-        if t.is_quantized or self.serializer == SERIALIZER_TORCH_SAVE:
-            # rare path: bring to host with torch, then pickle
+        if t.is_quantized or self.serializer != SERIALIZER_BUFFER:
+            # rare path: bring to host with torch, then pack
             cpu = t.cpu()
+            if self.serializer == SERIALIZER_QTENSOR:
+                return qtensor_as_bytes(cpu)
             return torch_save_as_bytes(cpu)
         if is_uvm_tensor(t):
             # managed memory is CPU-addressable: serialize zero-copy, no
@@ -302,6 +307,8 @@ class TensorBufferStager(BufferStager):
 
     def _stage_cpu(self, ctx: StageContext) -> BufferType:
         t = self.tensor.detach()
+        if self.serializer == SERIALIZER_QTENSOR:
+            return qtensor_as_bytes(t)
         if self.serializer == SERIALIZER_TORCH_SAVE:
             return torch_save_as_bytes(t)
         if self._should_copy_cpu_tensor(t, ctx):
@@ -365,8 +372,14 @@ class TensorBufferConsumer(BufferConsumer):
 
     async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
         def work() -> None:
-            dtype = str_to_dtype(self.entry.dtype) if not self.entry.serializer == SERIALIZER_TORCH_SAVE else None
-            if self.entry.serializer == SERIALIZER_TORCH_SAVE:
+            dtype = (
+                str_to_dtype(self.entry.dtype)
+                if self.entry.serializer == SERIALIZER_BUFFER
+                else None
+            )
+            if self.entry.serializer == SERIALIZER_QTENSOR:
+                loaded = qtensor_from_bytes(buf)
+            elif self.entry.serializer == SERIALIZER_TORCH_SAVE:
                 loaded = torch_load_from_bytes(bytes(buf))
             elif (
                 self._pinned_block is not None

@@ -74,12 +74,20 @@ def dtype_size_bytes(dtype_str: str) -> int:
 
 SERIALIZER_BUFFER = "buffer"
 SERIALIZER_TORCH_SAVE = "torch_save"
+SERIALIZER_QTENSOR = "qtensor"
 
 
 def pick_serializer(tensor: torch.Tensor) -> str:
-    """Raw-buffer serialization for every plain dtype; torch_save for
-    quantized tensors (their scale/zero-point layout is torch-internal)."""
+    """Raw-buffer serialization for every plain dtype; a compact custom
+    binary layout for quantized tensors (parity with reference
+    torchsnapshot/serialization.py:278-477); torch_save for anything
+    exotic."""
     if tensor.is_quantized:
+        if tensor.qscheme() in (
+            torch.per_tensor_affine,
+            torch.per_channel_affine,
+        ):
+            return SERIALIZER_QTENSOR
         return SERIALIZER_TORCH_SAVE
     if tensor.dtype in DTYPE_TO_STR:
         return SERIALIZER_BUFFER
@@ -120,6 +128,86 @@ def tensor_from_memoryview(
         return torch.empty(shape, dtype=dtype)
     u8 = torch.frombuffer(mv, dtype=torch.uint8)
     return u8.view(dtype).reshape(shape)
+
+
+# -- quantized-tensor serializer ---------------------------------------------
+#
+# Layout (little-endian):
+#   u32 header_len | header json | raw storage bytes | scales f64[] | zps i64[]
+# Per-tensor: one scale + one zero point. Per-channel: axis in the header,
+# C scales + C zero points. The storage bytes are the int representation
+# (int_repr), so the payload stays zero-copy-readable.
+
+
+def qtensor_as_bytes(tensor: torch.Tensor) -> bytes:
+    import json as _json
+    import struct as _struct
+
+    if tensor.qscheme() == torch.per_tensor_affine:
+        header = {
+            "scheme": "per_tensor",
+            "dtype": dtype_to_str(tensor.dtype),
+            "shape": list(tensor.shape),
+        }
+        scales = _struct.pack("<d", float(tensor.q_scale()))
+        zps = _struct.pack("<q", int(tensor.q_zero_point()))
+    elif tensor.qscheme() == torch.per_channel_affine:
+        header = {
+            "scheme": "per_channel",
+            "dtype": dtype_to_str(tensor.dtype),
+            "shape": list(tensor.shape),
+            "axis": int(tensor.q_per_channel_axis()),
+        }
+        s = tensor.q_per_channel_scales().to(torch.float64).contiguous()
+        z = tensor.q_per_channel_zero_points().to(torch.int64).contiguous()
+        scales = bytes(tensor_as_memoryview(s))
+        zps = bytes(tensor_as_memoryview(z))
+    else:
+        raise ValueError(f"unsupported qscheme: {tensor.qscheme()}")
+    hdr = _json.dumps(header).encode("utf-8")
+    storage = bytes(tensor_as_memoryview(tensor.int_repr().contiguous()))
+    return (
+        _struct.pack("<I", len(hdr)) + hdr + storage + scales + zps
+    )
+
+
+def qtensor_from_bytes(data) -> torch.Tensor:
+    import json as _json
+    import struct as _struct
+
+    mv = memoryview(data)
+    (hdr_len,) = _struct.unpack("<I", mv[:4])
+    header = _json.loads(bytes(mv[4 : 4 + hdr_len]).decode("utf-8"))
+    dtype = str_to_dtype(header["dtype"])
+    shape = tuple(header["shape"])
+    numel = 1
+    for s in shape:
+        numel *= s
+    # int_repr element size: qint8/quint8 -> 1, qint32 -> 4
+    int_dtype = {
+        torch.qint8: torch.int8,
+        torch.quint8: torch.uint8,
+        torch.qint32: torch.int32,
+    }[dtype]
+    storage_len = numel * int_dtype.itemsize
+    off = 4 + hdr_len
+    storage = tensor_from_memoryview(
+        mv[off : off + storage_len], int_dtype, shape
+    )
+    off += storage_len
+    if header["scheme"] == "per_tensor":
+        (scale,) = _struct.unpack("<d", mv[off : off + 8])
+        (zp,) = _struct.unpack("<q", mv[off + 8 : off + 16])
+        return torch._make_per_tensor_quantized_tensor(storage, scale, zp)
+    n_ch = shape[header["axis"]]
+    scales = tensor_from_memoryview(
+        mv[off : off + 8 * n_ch], torch.float64, (n_ch,)
+    )
+    off += 8 * n_ch
+    zps = tensor_from_memoryview(mv[off : off + 8 * n_ch], torch.int64, (n_ch,))
+    return torch._make_per_channel_quantized_tensor(
+        storage, scales, zps, header["axis"]
+    )
 
 
 # -- torch_save serializer ---------------------------------------------------
