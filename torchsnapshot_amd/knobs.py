@@ -50,7 +50,7 @@ def get_max_shard_size_bytes() -> int:
 
 def get_slab_size_threshold_bytes() -> int:
     """Write requests smaller than this are packed into batched slabs."""
-    return _env_bytes("TSAMD_SLAB_SIZE_THRESHOLD_BYTES", 256 * _MB)
+    return _env_bytes("TSAMD_SLAB_SIZE_THRESHOLD_BYTES", 512 * _MB)
 
 
 def is_batching_disabled() -> bool:
@@ -65,12 +65,12 @@ def is_partitioner_disabled() -> bool:
 
 def get_max_io_concurrency() -> int:
     """Maximum concurrent storage I/O operations per rank."""
-    return _env_int("TSAMD_MAX_PER_RANK_IO_CONCURRENCY", 16)
+    return _env_int("TSAMD_MAX_PER_RANK_IO_CONCURRENCY", 24)
 
 
 def get_num_staging_threads() -> int:
     """Executor threads used for CPU-side staging (GIL-releasing copies)."""
-    return _env_int("TSAMD_NUM_STAGING_THREADS", 4)
+    return _env_int("TSAMD_NUM_STAGING_THREADS", 8)
 
 
 def get_memory_budget_override_bytes() -> int | None:
@@ -116,7 +116,7 @@ def is_hip_staging_disabled() -> bool:
 
 def get_storage_write_chunk_bytes() -> int:
     """Chunk size for filesystem pwrite calls (large sequential writes)."""
-    return _env_bytes("TSAMD_FS_WRITE_CHUNK_BYTES", 64 * _MB)
+    return _env_bytes("TSAMD_FS_WRITE_CHUNK_BYTES", 256 * _MB)
 
 
 # -- elasticity --------------------------------------------------------------

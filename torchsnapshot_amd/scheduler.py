@@ -156,6 +156,16 @@ class PendingIOWork:
         """Block until all storage I/O finished; re-raise pipeline errors."""
         self._thread.join()
         self._maybe_raise()
+        import os
+
+        if os.environ.get("TSAMD_TIMING") and self.stats.staged_bytes:
+            s = self.stats
+            wall = (s.end_ts or 0) - s.begin_ts
+            print(
+                f"[tsamd timing] pipeline: wall={wall:.2f}s reqs={s.total_reqs} "
+                f"staged={s.staged_bytes/1e9:.2f}GB stage_busy={s.stage_s:.2f}s "
+                f"io_busy={s.io_s:.2f}s consume_busy={s.consume_s:.2f}s"
+            )
 
     def done(self) -> bool:
         return self._done_event.is_set()
