@@ -9,6 +9,14 @@ config (48 layers, hidden 7168): each rank holds 1/world_size of the
 fp16 params and fp32 optimizer moments as flat tensors.
 """
 
+import os
+import sys
+
+sys.path.insert(
+    0, os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+)
+
+
 import argparse
 import os
 import shutil

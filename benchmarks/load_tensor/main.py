@@ -5,6 +5,14 @@ with and without a 100 MB memory budget, reporting wall time and peak
 RSS delta — demonstrating that tiled byte-range reads bound host memory.
 """
 
+import os
+import sys
+
+sys.path.insert(
+    0, os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+)
+
+
 import argparse
 import os
 import shutil

@@ -10,6 +10,14 @@ Launch: python -m torch.distributed.run --nproc-per-node N \
             --master-addr 127.0.0.1 benchmarks/sharded_embedding/main.py
 """
 
+import os
+import sys
+
+sys.path.insert(
+    0, os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+)
+
+
 import argparse
 import os
 import shutil

@@ -7,6 +7,12 @@ Launch: python -m torch.distributed.run --nproc-per-node 2 \
 """
 
 import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+import os
 import tempfile
 
 import torch

@@ -1,6 +1,12 @@
 #!/usr/bin/env python3
 """Minimal example: checkpoint a model + optimizer + progress, resume."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import tempfile
 
 import torch

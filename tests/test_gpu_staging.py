@@ -196,3 +196,46 @@ def test_uvm_managed_tensor():
         out = StateDict(emb=torch.zeros(128, 64))
         snap.restore({"sd": out})
         assert torch.equal(out["emb"], t)
+
+
+def test_dtensor_on_gpu_single_rank():
+    """DTensor save/restore on a CUDA 1-rank mesh (RCCL world of 1) —
+    exercises the DTensor preparer through the HIP staging engine."""
+    import torch.distributed as dist
+
+    created = False
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+        created = True
+    try:
+        from torch.distributed.device_mesh import init_device_mesh
+        from torch.distributed.tensor import distribute_tensor
+        from torch.distributed.tensor.placement_types import Shard
+
+        mesh = init_device_mesh("cuda", (1,))
+        full = torch.rand(512, 64, device="cuda")
+        dt = distribute_tensor(full, mesh, [Shard(0)])
+
+        class Holder:
+            def __init__(self, dt):
+                self.dt = dt
+
+            def state_dict(self):
+                return {"dt": self.dt}
+
+            def load_state_dict(self, sd):
+                self.dt = sd["dt"]
+
+        with tmp_snapshot_path() as path:
+            snap = Snapshot.take(path, {"obj": Holder(dt)})
+            dt2 = distribute_tensor(
+                torch.zeros(512, 64, device="cuda"), mesh, [Shard(0)]
+            )
+            holder = Holder(dt2)
+            snap.restore({"obj": holder})
+            assert torch.equal(holder.dt.to_local(), full)
+    finally:
+        if created:
+            dist.destroy_process_group()

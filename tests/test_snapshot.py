@@ -201,3 +201,26 @@ def test_get_state_dict_for_key():
         snapshot = Snapshot.take(path, {"model": model})
         sd = snapshot.get_state_dict_for_key("model")
         assert check_state_dict_eq(dict(model.state_dict()), dict(sd))
+
+
+def test_restore_strict_false():
+    """A snapshot saved from a smaller module restores into a bigger one
+    with strict=False (missing keys keep their init values)."""
+    small = torch.nn.Linear(4, 4)
+
+    class Bigger(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.weight = torch.nn.Parameter(torch.zeros(4, 4))
+            self.bias = torch.nn.Parameter(torch.zeros(4))
+            self.extra = torch.nn.Parameter(torch.ones(3))
+
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"m": small})
+        big = Bigger()
+        snapshot.restore({"m": big}, strict=False)
+        assert torch.equal(big.weight.data, small.weight.data)
+        assert torch.equal(big.bias.data, small.bias.data)
+        assert torch.equal(big.extra.data, torch.ones(3))
+        with pytest.raises(RuntimeError):
+            snapshot.restore({"m": Bigger()}, strict=True)

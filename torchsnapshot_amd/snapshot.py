@@ -381,7 +381,10 @@ class Snapshot:
 
     # -- load ---------------------------------------------------------------
 
-    def restore(self, app_state: AppState) -> None:
+    def restore(self, app_state: AppState, strict: Optional[bool] = None) -> None:
+        """In-place restore. ``strict`` (when not None) is forwarded to
+        load_state_dict for statefuls that accept it (nn.Module etc.), so
+        partially-matching snapshots can restore with strict=False."""
         torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.restore")
         self._validate_app_state(app_state)
         pg_wrapper = PGWrapper(self.pg)
@@ -411,6 +414,7 @@ class Snapshot:
                         payload_entries=payload_entries,
                         storage=storage,
                         pg_wrapper=pg_wrapper,
+                        strict=strict,
                     )
                     pg_wrapper.barrier()
             finally:
@@ -428,6 +432,7 @@ class Snapshot:
         payload_entries: Dict[str, Entry],
         storage: StoragePlugin,
         pg_wrapper: PGWrapper,
+        strict: Optional[bool] = None,
     ) -> None:
         if stateful is None:
             return
@@ -471,6 +476,12 @@ class Snapshot:
         )
         values = {p: f.obj for p, f in futs.items()}
         state_dict_to_load = inflate(sub_manifest, values, prefix=key)
+        if strict is not None:
+            try:
+                stateful.load_state_dict(state_dict_to_load, strict=strict)
+                return
+            except TypeError:
+                pass  # stateful doesn't accept strict
         stateful.load_state_dict(state_dict_to_load)
 
     def get_state_dict_for_key(self, key: str) -> Dict[str, Any]:
