@@ -31,6 +31,7 @@ import psutil
 from . import knobs
 from .io_types import ReadIO, ReadReq, StageContext, StoragePlugin, WriteIO, WriteReq
 from .pg_wrapper import PGWrapper
+from .roctx import roctx_range
 
 logger = logging.getLogger(__name__)
 
@@ -238,7 +239,8 @@ def execute_write_reqs(
             try:
                 async with staging_sem:
                     t0 = time.monotonic()
-                    buf = await req.stager.stage_buffer(ctx)
+                    with roctx_range(f"tsamd:stage:{req.path}"):
+                        buf = await req.stager.stage_buffer(ctx)
                     stats.stage_s += time.monotonic() - t0
                 nbytes = memoryview(buf).nbytes
                 stats.staged_reqs += 1
@@ -248,7 +250,8 @@ def execute_write_reqs(
                     all_staged.set()
                 async with io_sem:
                     t0 = time.monotonic()
-                    await storage.write(WriteIO(path=req.path, buf=buf))
+                    with roctx_range(f"tsamd:write:{req.path}"):
+                        await storage.write(WriteIO(path=req.path, buf=buf))
                     stats.io_s += time.monotonic() - t0
                 stats.io_bytes += nbytes
                 stats.done_reqs += 1
