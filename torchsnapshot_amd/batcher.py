@@ -222,9 +222,67 @@ class BatchedBufferConsumer(BufferConsumer):
             total += m.consumer.get_consuming_cost_bytes()
         return total
 
-    async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
-        mv = memoryview(buf)
+    def _device_fast_path_target(self):
+        """If every member is a buffer-serialized tensor restore onto the
+        same CUDA device, the whole span can go up in ONE H2D and be sliced
+        on the GPU — per-member H2D round trips dominate otherwise
+        (measured 0.43 GB/s vs multi-GB/s for 2000 small tensors)."""
+        from .io_preparers.tensor import TensorBufferConsumer
+
+        device = None
         for m in self.members:
-            s, e = m.byte_range
-            sub = mv[s - self.span_start : e - self.span_start]
-            await m.consumer.consume_buffer(ctx, sub)
+            c = m.consumer
+            if (
+                not isinstance(c, TensorBufferConsumer)
+                or c.tensor_out is None
+                or c.tensor_out.device.type != "cuda"
+                or c.entry.serializer != SERIALIZER_BUFFER
+            ):
+                return None
+            if device is None:
+                device = c.tensor_out.device
+            elif c.tensor_out.device != device:
+                return None
+        return device
+
+    async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
+        device = self._device_fast_path_target()
+        if device is None:
+            mv = memoryview(buf)
+            for m in self.members:
+                s, e = m.byte_range
+                sub = mv[s - self.span_start : e - self.span_start]
+                await m.consumer.consume_buffer(ctx, sub)
+            return
+
+        def work() -> None:
+            import asyncio as _  # noqa: F401
+
+            from .io_preparers.tensor import tensor_copy
+            from .ops.staging import get_pinned_pool
+            from .serialization import str_to_dtype
+
+            mv = memoryview(buf)
+            nbytes = mv.nbytes
+            pool = get_pinned_pool()
+            block = pool.acquire(max(nbytes, 1))
+            try:
+                src = torch.frombuffer(mv, dtype=torch.uint8)
+                block.tensor[:nbytes].copy_(src)
+                dev_span = block.tensor[:nbytes].to(device, non_blocking=False)
+            finally:
+                pool.release(block)
+            for m in self.members:
+                c = m.consumer
+                s, e = m.byte_range
+                sub = dev_span[s - self.span_start : e - self.span_start]
+                dtype = str_to_dtype(c.entry.dtype)
+                loaded = (
+                    sub.view(dtype).reshape(tuple(c.entry.shape))
+                    if dtype != torch.uint8
+                    else sub.reshape(tuple(c.entry.shape))
+                )
+                tensor_copy(c.tensor_out, loaded)
+                c.fut.obj = c.tensor_out
+
+        await asyncio.get_running_loop().run_in_executor(ctx.executor, work)
