@@ -1,8 +1,5 @@
 """Unit tests for load-path manifest transformations."""
 
-import pytest
-import torch
-
 from torchsnapshot_amd.flatten import inflate
 from torchsnapshot_amd.manifest import (
     ChunkedTensorEntry,

@@ -3,7 +3,6 @@ import torch
 
 from torchsnapshot_amd.serialization import (
     SERIALIZER_BUFFER,
-    SERIALIZER_TORCH_SAVE,
     dtype_to_str,
     pick_serializer,
     str_to_dtype,

@@ -11,7 +11,6 @@ import torch
 
 from torchsnapshot_amd.io_types import StageContext
 from torchsnapshot_amd.io_preparers.sharded_tensor import (
-    Overlap,
     compute_overlap,
     narrow_nd,
     plan_shard_reads,

@@ -17,7 +17,7 @@ import pickle
 import socket
 import time
 from datetime import timedelta
-from typing import Any, List, Optional
+from typing import Any, List
 
 import torch.distributed as dist
 

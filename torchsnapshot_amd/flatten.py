@@ -25,7 +25,6 @@ from typing import Any, Dict, List, Tuple, Union
 
 from .manifest import (
     DictEntry,
-    Entry,
     ListEntry,
     Manifest,
     OrderedDictEntry,

@@ -16,7 +16,6 @@ from typing import Any, List, Optional, Tuple
 
 import torch
 
-from . import knobs
 from .io_types import ReadReq, WriteReq
 from .manifest import (
     ChunkedTensorEntry,

@@ -22,7 +22,6 @@ from __future__ import annotations
 
 import base64
 import json
-import struct
 from dataclasses import dataclass, field
 from typing import Any, Dict, List, Optional, Union
 

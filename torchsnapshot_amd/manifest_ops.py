@@ -31,7 +31,6 @@ from .manifest import (
     is_container_entry,
 )
 from .manifest_utils import (
-    get_replicated_ranks,
     is_fully_replicated_entry,
     is_sharded_entry,
 )

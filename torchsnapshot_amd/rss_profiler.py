@@ -9,7 +9,6 @@ from __future__ import annotations
 import contextlib
 import threading
 import time
-from collections import deque
 from typing import Deque, Iterator
 
 import psutil

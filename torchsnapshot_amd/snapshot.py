@@ -44,7 +44,6 @@ from .manifest import (
     METADATA_FILENAME,
     PrimitiveEntry,
     SnapshotMetadata,
-    is_container_entry,
 )
 from .manifest_ops import (
     get_manifest_for_rank,
@@ -59,7 +58,6 @@ from .scheduler import (
     execute_write_reqs,
     get_process_memory_budget_bytes,
     sync_execute_read_reqs,
-    sync_execute_write_reqs,
 )
 from .stateful import AppState, Stateful
 from .storage import url_to_storage_plugin

@@ -20,7 +20,7 @@ import hashlib
 import hmac
 import os
 import urllib.parse
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional
 
 from ..io_types import ReadIO, StoragePlugin, WriteIO
 
