@@ -32,6 +32,7 @@ from . import knobs
 from .io_types import ReadIO, ReadReq, StageContext, StoragePlugin, WriteIO, WriteReq
 from .pg_wrapper import PGWrapper
 from .roctx import roctx_range
+from . import integrity
 
 logger = logging.getLogger(__name__)
 
@@ -129,12 +130,15 @@ class PendingIOWork:
         done_event: threading.Event,
         stats: ExecutionStats,
         exc_holder: List[BaseException],
+        checksums: Optional[dict] = None,
     ) -> None:
         self._thread = thread
         self._staged_event = staged_event
         self._done_event = done_event
         self.stats = stats
         self._exc_holder = exc_holder
+        # {payload_path: xxh3 hex}, filled when TSAMD_CHECKSUM=1
+        self.checksums = checksums if checksums is not None else {}
 
     def _maybe_raise(self) -> None:
         if self._exc_holder:
@@ -177,6 +181,7 @@ def _spawn_pipeline(
     stats: ExecutionStats,
     staged_event: threading.Event,
     done_event: threading.Event,
+    checksums: Optional[dict] = None,
 ) -> PendingIOWork:
     exc_holder: List[BaseException] = []
 
@@ -194,7 +199,9 @@ def _spawn_pipeline(
         target=runner, name="tsamd-io-pipeline", daemon=True
     )
     thread.start()
-    return PendingIOWork(thread, staged_event, done_event, stats, exc_holder)
+    return PendingIOWork(
+        thread, staged_event, done_event, stats, exc_holder, checksums
+    )
 
 
 # ---------------------------------------------------------------------------
@@ -212,6 +219,8 @@ def execute_write_reqs(
     stats = ExecutionStats(total_reqs=len(write_reqs))
     staged_event = threading.Event()
     done_event = threading.Event()
+    checksums: dict = {}
+    do_checksum = integrity.checksumming_enabled()
 
     # Largest first: big buffers claim budget early, small ones fill gaps.
     ordered = sorted(
@@ -243,6 +252,9 @@ def execute_write_reqs(
                         buf = await req.stager.stage_buffer(ctx)
                     stats.stage_s += time.monotonic() - t0
                 nbytes = memoryview(buf).nbytes
+                if do_checksum:
+                    checksums[req.path] = await asyncio.get_running_loop(
+                    ).run_in_executor(executor, integrity.hash_buffer, buf)
                 stats.staged_reqs += 1
                 stats.staged_bytes += nbytes
                 staged_remaining -= 1
@@ -278,7 +290,7 @@ def execute_write_reqs(
             reporter.cancel()
             executor.shutdown(wait=False)
 
-    return _spawn_pipeline(main, stats, staged_event, done_event)
+    return _spawn_pipeline(main, stats, staged_event, done_event, checksums)
 
 
 def sync_execute_write_reqs(
@@ -321,6 +333,7 @@ def execute_read_reqs(
     storage: StoragePlugin,
     memory_budget_bytes: int,
     rank: int,
+    checksums: Optional[dict] = None,
 ) -> PendingIOWork:
     stats = ExecutionStats(total_reqs=len(read_reqs))
     staged_event = threading.Event()
@@ -356,6 +369,11 @@ def execute_read_reqs(
                     stats.io_s += time.monotonic() - t0
                 buf = read_io.buf
                 stats.io_bytes += memoryview(buf).nbytes
+                if checksums and req.byte_range is None:
+                    await asyncio.get_running_loop().run_in_executor(
+                        executor, integrity.verify_buffer, req.path, buf,
+                        checksums,
+                    )
                 t0 = time.monotonic()
                 await req.consumer.consume_buffer(ctx, buf)
                 stats.consume_s += time.monotonic() - t0
@@ -383,8 +401,11 @@ def sync_execute_read_reqs(
     storage: StoragePlugin,
     memory_budget_bytes: int,
     rank: int,
+    checksums: Optional[dict] = None,
 ) -> ExecutionStats:
-    pending = execute_read_reqs(read_reqs, storage, memory_budget_bytes, rank)
+    pending = execute_read_reqs(
+        read_reqs, storage, memory_budget_bytes, rank, checksums
+    )
     pending.complete()
     import os
 

@@ -170,6 +170,11 @@ class Snapshot:
                     is_async=False,
                 )
                 pending_io_work.complete()
+                from .integrity import write_checksum_file
+
+                write_checksum_file(
+                    storage, pg_wrapper.get_rank(), pending_io_work.checksums
+                )
                 cls._commit(storage, metadata, pg_wrapper)
             finally:
                 storage.sync_close()
@@ -399,6 +404,13 @@ class Snapshot:
                 rank_manifest, payload_entries = get_manifest_for_rank(
                     metadata, pg_wrapper.get_rank()
                 )
+                from . import integrity
+
+                checksums = (
+                    integrity.load_checksums(storage, metadata.world_size)
+                    if integrity.verification_enabled()
+                    else None
+                )
                 all_keys = self._gather_keys(app_state, pg_wrapper)
                 # RNG states restore last so nothing after perturbs them
                 all_keys.sort(
@@ -413,6 +425,7 @@ class Snapshot:
                         storage=storage,
                         pg_wrapper=pg_wrapper,
                         strict=strict,
+                        checksums=checksums,
                     )
                     pg_wrapper.barrier()
             finally:
@@ -431,6 +444,7 @@ class Snapshot:
         storage: StoragePlugin,
         pg_wrapper: PGWrapper,
         strict: Optional[bool] = None,
+        checksums: Optional[Dict[str, str]] = None,
     ) -> None:
         if stateful is None:
             return
@@ -470,7 +484,8 @@ class Snapshot:
         read_reqs = _batch_reads(read_reqs)
         budget = get_process_memory_budget_bytes(pg_wrapper)
         sync_execute_read_reqs(
-            read_reqs, storage, budget, rank=pg_wrapper.get_rank()
+            read_reqs, storage, budget, rank=pg_wrapper.get_rank(),
+            checksums=checksums,
         )
         values = {p: f.obj for p, f in futs.items()}
         state_dict_to_load = inflate(sub_manifest, values, prefix=key)
@@ -736,6 +751,13 @@ class PendingSnapshot:
     def _complete(self) -> None:
         try:
             self._pending_io_work.complete()
+            from .integrity import write_checksum_file
+
+            write_checksum_file(
+                self._storage,
+                self._pg_wrapper.get_rank(),
+                self._pending_io_work.checksums,
+            )
             self._barrier.arrive()
             if self._pg_wrapper.get_rank() == 0:
                 self._storage.sync_write(
