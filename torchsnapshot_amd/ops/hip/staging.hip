@@ -328,8 +328,18 @@ long long launch_pack(const std::vector<unsigned long long>& flat, int n,
                              hipMemcpyHostToDevice, stream));
   }
 
+  // Grid cap. Direct mode is PCIe-bound: 128 workgroups already sustain
+  // the measured ~51 GB/s host-link ceiling (grid sweep, profiles/), so a
+  // small grid leaves the CUs to overlapped training compute. Slab mode
+  // gathers at HBM speed and wants the chip filled (all 8 XCDs).
+  bool direct_mode = (slab_ptr == 0);
+  unsigned long long grid_cap = direct_mode ? 256ull : 16384ull;
+  if (const char* env = getenv("TSAMD_PACK_GRID")) {
+    long v = atol(env);
+    if (v > 0) grid_cap = (unsigned long long)v;
+  }
   unsigned int grid = (unsigned int)std::min<unsigned long long>(
-      total_wus == 0 ? 1 : total_wus, 16384ull);
+      total_wus == 0 ? 1 : total_wus, grid_cap);
   if (gather) {
     hipLaunchKernelGGL(pack_kernel<true>, dim3(grid), dim3(kBlockThreads), 0,
                        stream, d_descs, n, d_prefix, total_wus, flat_base);
