@@ -207,7 +207,9 @@ def test_dtensor_on_gpu_single_rank():
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29531")
-        dist.init_process_group("nccl", rank=0, world_size=1)
+        dist.init_process_group(
+            "nccl", rank=0, world_size=1, device_id=torch.device("cuda", 0)
+        )
         created = True
     try:
         from torch.distributed.device_mesh import init_device_mesh

@@ -224,3 +224,18 @@ def test_restore_strict_false():
         assert torch.equal(big.extra.data, torch.ones(3))
         with pytest.raises(RuntimeError):
             snapshot.restore({"m": Bigger()}, strict=True)
+
+
+def test_empty_app_state():
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {})
+        assert snapshot.get_manifest() == {}
+        snapshot.restore({})
+
+
+def test_invalid_app_state_keys():
+    with tmp_snapshot_path() as path:
+        with pytest.raises(ValueError, match="invalid"):
+            Snapshot.take(path, {"bad/key": StateDict(a=1)})
+        with pytest.raises(TypeError):
+            Snapshot.take(path, {"x": object()})

@@ -582,6 +582,11 @@ class Snapshot:
                 f"app_state must be Dict[str, Stateful], got {type(app_state)}"
             )
         for key, value in app_state.items():
+            if not isinstance(key, str) or "/" in key:
+                raise ValueError(
+                    f"app_state key {key!r} is invalid: keys must be strings "
+                    "without '/'"
+                )
             if not isinstance(value, Stateful):
                 raise TypeError(
                     f"app_state['{key}'] ({type(value).__name__}) does not "
