@@ -93,3 +93,34 @@ def test_failed_sync_take_commits_no_metadata():
             with pytest.raises(RuntimeError):
                 Snapshot.take(path, {"sd": sd})
         assert not os.path.exists(os.path.join(path, ".snapshot_metadata"))
+
+
+def _async_take_dist(tmpdir: str) -> None:
+    import os
+
+    import torch.distributed as dist
+
+    from torchsnapshot_amd.test_utils import check_state_dict_eq
+
+    torch.manual_seed(10 + dist.get_rank())
+    sd = StateDict(
+        mine=torch.rand(64, 32),
+        shared=torch.full((16,), float(dist.get_rank())),
+    )
+    path = os.path.join(tmpdir, "snap")
+    pending = Snapshot.async_take(path, {"sd": sd})
+    snapshot = pending.wait()
+    assert os.path.exists(os.path.join(path, ".snapshot_metadata"))
+    out = StateDict(mine=torch.zeros(64, 32), shared=torch.zeros(16))
+    snapshot.restore({"sd": out})
+    assert torch.equal(out["mine"], sd["mine"])
+    assert torch.equal(out["shared"], sd["shared"])
+
+
+def test_async_take_world2():
+    import tempfile as tf
+
+    from torchsnapshot_amd.test_utils import run_multiprocess
+
+    with tf.TemporaryDirectory() as d:
+        run_multiprocess(2, _async_take_dist, d)
