@@ -324,10 +324,15 @@ class StagingEngine:
         if total == 0:
             batch._done = True
             return batch
-        if self._use_ext:
-            self._stage_ext(tensors, items, batch, total)
-        else:
-            self._stage_torch_fallback(tensors, items, batch)
+        try:
+            if self._use_ext:
+                self._stage_ext(tensors, items, batch, total)
+            else:
+                self._stage_torch_fallback(tensors, items, batch)
+        except BaseException:
+            # don't leak the pool slot on launch failure
+            get_pinned_pool().release(pinned)
+            raise
         return batch
 
     # -- native path --------------------------------------------------------

@@ -267,8 +267,13 @@ def execute_write_reqs(
                     stats.io_s += time.monotonic() - t0
                 stats.io_bytes += nbytes
                 stats.done_reqs += 1
-                req.stager.release_buffer()
             finally:
+                # release pooled staging buffers on success AND failure so
+                # a failed snapshot doesn't starve the pinned pool
+                try:
+                    req.stager.release_buffer()
+                except Exception:
+                    logger.exception("release_buffer failed for %s", req.path)
                 await budget.release(cost)
 
         tasks = [asyncio.create_task(handle(r)) for r in ordered]
