@@ -217,6 +217,31 @@ class S3StoragePlugin(StoragePlugin):
             if resp.status not in (200, 204, 404):
                 raise RuntimeError(f"S3 DELETE {path}: {resp.status}")
 
+    # sync one-shot ops (metadata reads etc.) run on throwaway event
+    # loops; close the per-loop session inside the same loop so aiohttp
+    # connections never leak
+    def sync_read(self, read_io) -> None:
+        from ..scheduler import run_coro_sync
+
+        async def go():
+            try:
+                await self.read(read_io)
+            finally:
+                await self.close()
+
+        run_coro_sync(go())
+
+    def sync_write(self, write_io) -> None:
+        from ..scheduler import run_coro_sync
+
+        async def go():
+            try:
+                await self.write(write_io)
+            finally:
+                await self.close()
+
+        run_coro_sync(go())
+
     async def close(self) -> None:
         sess = self._sessions.pop(id(asyncio.get_running_loop()), None)
         if sess is not None and not sess.closed:
