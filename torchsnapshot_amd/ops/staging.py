@@ -224,6 +224,20 @@ class PinnedPool:
             self._free.append(block.tensor)
             self._cond.notify()
 
+    def prealloc_one(self) -> bool:
+        """Allocate one new block straight into the free list (never takes
+        from it), so warming can proceed concurrently with real staging
+        without hoarding blocks."""
+        with self._cond:
+            if self._allocated >= self.max_blocks:
+                return False
+            self._allocated += 1
+        tensor = torch.empty(self.block_size, dtype=torch.uint8, pin_memory=True)
+        with self._cond:
+            self._free.append(tensor)
+            self._cond.notify()
+        return True
+
 
 _pool_lock = threading.Lock()
 _pool: Optional[PinnedPool] = None
@@ -235,6 +249,37 @@ def get_pinned_pool() -> PinnedPool:
         if _pool is None:
             _pool = PinnedPool()
         return _pool
+
+
+def warm_pinned_pool(nbytes: Optional[int] = None, background: bool = True) -> None:
+    """Pre-allocate (and so pre-register) pinned blocks. Pinning is a
+    one-time ~GB/s kernel-side cost; warming it off the critical path
+    keeps the FIRST checkpoint as fast as the rest. Called automatically
+    (in the background) when a StagingEngine is first created."""
+    pool = get_pinned_pool()
+
+    def work() -> None:
+        import os
+
+        if nbytes is not None:
+            target = nbytes
+        else:
+            # default: don't pin the whole pool for workloads that may
+            # never need it — 8 GB covers most first checkpoints
+            target = min(
+                pool.block_size * pool.max_blocks,
+                int(float(os.environ.get("TSAMD_POOL_WARM_BYTES", 8 * 1024**3))),
+            )
+        allocated = 0
+        while allocated < target and pool.prealloc_one():
+            allocated += pool.block_size
+
+    if background:
+        threading.Thread(
+            target=work, name="tsamd-pool-warm", daemon=True
+        ).start()
+    else:
+        work()
 
 
 # ---------------------------------------------------------------------------
@@ -289,6 +334,9 @@ class StagingEngine:
         self._use_ext = HIP_EXT_AVAILABLE and not knobs.is_hip_staging_disabled()
         if not self._use_ext and not knobs.is_hip_staging_disabled():
             _require_ext()
+        # hide the one-time pinned-registration cost behind whatever runs
+        # before the first checkpoint
+        warm_pinned_pool()
 
     def stage(self, tensors: Sequence[torch.Tensor]) -> StagedBatch:
         """Start async D2H staging of device tensors into one pinned slab.
