@@ -124,3 +124,21 @@ def test_async_take_world2():
 
     with tf.TemporaryDirectory() as d:
         run_multiprocess(2, _async_take_dist, d)
+
+
+def test_two_concurrent_async_takes():
+    """Two async snapshots of different app states to different paths may
+    overlap; both must commit correctly."""
+    sd1 = StateDict(a=torch.rand(128, 64))
+    sd2 = StateDict(b=torch.rand(256, 16))
+    with tempfile.TemporaryDirectory() as d:
+        with _patch_plugin(SlowFSStoragePlugin):
+            p1 = Snapshot.async_take(os.path.join(d, "s1"), {"sd": sd1})
+            p2 = Snapshot.async_take(os.path.join(d, "s2"), {"sd": sd2})
+            s1 = p1.wait()
+            s2 = p2.wait()
+        out1, out2 = StateDict(), StateDict()
+        s1.restore({"sd": out1})
+        s2.restore({"sd": out2})
+        assert torch.equal(out1["a"], sd1["a"])
+        assert torch.equal(out2["b"], sd2["b"])
