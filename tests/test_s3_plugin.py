@@ -39,6 +39,21 @@ class FakeS3:
             if request.method == "PUT":
                 self.objects[key] = await request.read()
                 return web.Response(status=200)
+            if request.method == "GET" and "list-type" in request.query:
+                prefix = request.query.get("prefix", "")
+                # path-style: request path is the bucket; stored keys are
+                # "<bucket>/<key>" but real S3 lists keys without bucket
+                bucket = key.split("/")[0] if key else ""
+                in_bucket = [
+                    k[len(bucket) + 1 :]
+                    for k in self.objects
+                    if k.startswith(bucket + "/")
+                ]
+                keys = [k for k in in_bucket if k.startswith(prefix)]
+                body = "<ListBucketResult>" + "".join(
+                    f"<Key>{k}</Key>" for k in sorted(keys)
+                ) + "</ListBucketResult>"
+                return web.Response(status=200, text=body)
             if request.method == "GET":
                 if key not in self.objects:
                     return web.Response(status=404)
@@ -140,3 +155,13 @@ def test_s3_missing_creds():
         for k, v in env_backup.items():
             if v is not None:
                 os.environ[k] = v
+
+
+def test_s3_delete_snapshot(fake_s3):
+    sd = StateDict(w=torch.rand(32))
+    snap = Snapshot.take(
+        "s3://bkt/del", {"sd": sd}, storage_options=_options(fake_s3)
+    )
+    assert any(k.startswith("bkt/del/") for k in fake_s3.objects)
+    snap.delete()
+    assert not any(k.startswith("bkt/del/") for k in fake_s3.objects)

@@ -239,3 +239,14 @@ def test_invalid_app_state_keys():
             Snapshot.take(path, {"bad/key": StateDict(a=1)})
         with pytest.raises(TypeError):
             Snapshot.take(path, {"x": object()})
+
+
+def test_snapshot_delete_fs():
+    sd = StateDict(a=torch.rand(8))
+    with tmp_snapshot_path() as path:
+        snapshot = Snapshot.take(path, {"sd": sd})
+        assert os.path.exists(path)
+        snapshot.delete()
+        assert not os.path.exists(path)
+        with pytest.raises(RuntimeError):
+            _ = Snapshot(path).metadata

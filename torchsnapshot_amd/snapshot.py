@@ -138,6 +138,25 @@ class Snapshot:
         """A deep copy of the global manifest as plain dicts."""
         return {k: v.to_dict() for k, v in self.metadata.manifest.items()}
 
+    def delete(self) -> None:
+        """Delete the snapshot (metadata first, so a concurrent reader sees
+        an invalid snapshot rather than a partially-deleted one)."""
+        storage = url_to_storage_plugin(self.path, self._storage_options)
+        try:
+            from .scheduler import run_coro_sync
+
+            async def go():
+                try:
+                    await storage.delete(METADATA_FILENAME)
+                except FileNotFoundError:
+                    pass
+                await storage.delete_dir("")
+
+            run_coro_sync(go())
+        finally:
+            storage.sync_close()
+        self._metadata = None
+
     # -- save ---------------------------------------------------------------
 
     @classmethod

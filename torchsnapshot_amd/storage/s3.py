@@ -209,6 +209,44 @@ class S3StoragePlugin(StoragePlugin):
             await asyncio.sleep(0.5 * 2**attempt)
         raise RuntimeError(f"S3 GET {read_io.path}: retries exhausted")
 
+    async def list_keys(self, prefix: str = "") -> list:
+        """ListObjectsV2 under the plugin root + prefix."""
+        import re
+
+        full_prefix = f"{self.prefix}/{prefix}" if self.prefix else prefix
+        keys = []
+        token = None
+        while True:
+            q = f"list-type=2&prefix={urllib.parse.quote(full_prefix, safe='')}"
+            if token:
+                q += f"&continuation-token={urllib.parse.quote(token, safe='')}"
+            base = self.endpoint.rstrip("/")
+            url = (
+                f"{base}/{self.bucket}?{q}"
+                if self._path_style
+                else f"{base}/?{q}"
+            )
+            headers = self.signer.sign("GET", url, _EMPTY_SHA256)
+            sess = await self._session()
+            async with sess.get(url, headers=headers) as resp:
+                if resp.status != 200:
+                    raise RuntimeError(f"S3 LIST failed: {resp.status}")
+                body = await resp.text()
+            keys += re.findall(r"<Key>([^<]+)</Key>", body)
+            m = re.search(
+                r"<NextContinuationToken>([^<]+)</NextContinuationToken>", body
+            )
+            if not m:
+                break
+            token = m.group(1)
+        strip = f"{self.prefix}/" if self.prefix else ""
+        return [k[len(strip):] if k.startswith(strip) else k for k in keys]
+
+    async def delete_dir(self, path: str) -> None:
+        prefix = f"{path}/" if path else ""
+        for key in await self.list_keys(prefix):
+            await self.delete(key)
+
     async def delete(self, path: str) -> None:
         url = self._url(path)
         headers = self.signer.sign("DELETE", url, _EMPTY_SHA256)
