@@ -1,0 +1,92 @@
+"""Property-based tests (hypothesis): flatten/inflate round-trips over
+arbitrary nestings, and pack-layout logical-order over random views."""
+
+import hypothesis.strategies as st
+import torch
+from hypothesis import given, settings
+
+from torchsnapshot_amd.flatten import flatten, inflate
+from torchsnapshot_amd.ops.staging import build_pack_items, collapse_layout
+
+# -- flatten/inflate ---------------------------------------------------------
+
+_keys = st.one_of(
+    st.text(
+        alphabet=st.characters(
+            blacklist_categories=("Cs",), blacklist_characters="\x00"
+        ),
+        max_size=8,
+    ),
+    st.integers(min_value=-3, max_value=99),
+)
+_leaves = st.one_of(
+    st.integers(min_value=-(2**40), max_value=2**40),
+    st.floats(allow_nan=False),
+    st.booleans(),
+    st.text(max_size=6),
+    st.binary(max_size=6),
+)
+
+
+def _nested(depth: int):
+    if depth == 0:
+        return _leaves
+    sub = _nested(depth - 1)
+    return st.one_of(
+        _leaves,
+        st.lists(sub, max_size=4),
+        st.dictionaries(_keys, sub, max_size=4),
+    )
+
+
+@settings(max_examples=80, deadline=None)
+@given(obj=st.dictionaries(st.text(max_size=6), _nested(3), max_size=4))
+def test_flatten_inflate_round_trip(obj):
+    manifest, flattened = flatten(obj, prefix="root")
+    rebuilt = inflate(manifest, flattened, prefix="root")
+    assert rebuilt == obj
+
+
+# -- pack layout -------------------------------------------------------------
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    dims=st.lists(st.integers(min_value=1, max_value=7), min_size=1, max_size=4),
+    data=st.data(),
+)
+def test_collapse_layout_any_view(dims, data):
+    t = torch.arange(int(torch.tensor(dims).prod()), dtype=torch.float32).reshape(
+        dims
+    )
+    # random slicing per dim
+    view = t
+    for d in range(t.dim()):
+        size = view.shape[d]
+        start = data.draw(st.integers(0, max(size - 1, 0)))
+        step = data.draw(st.integers(1, 3))
+        view = view.narrow(d, start, size - start)
+        idx = [slice(None)] * view.dim()
+        idx[d] = slice(None, None, step)
+        view = view[tuple(idx)]
+    maybe_permute = data.draw(st.booleans())
+    if maybe_permute and view.dim() > 1:
+        perm = data.draw(st.permutations(list(range(view.dim()))))
+        view = view.permute(perm)
+
+    sizes, strides = collapse_layout(view)
+    v2 = torch.as_strided(view, sizes, strides, view.storage_offset())
+    assert torch.equal(
+        v2.contiguous().reshape(-1), view.contiguous().reshape(-1)
+    )
+    # descriptor invariants
+    items, offsets, total = build_pack_items([view])
+    it = items[0]
+    if it.nbytes:
+        assert it.row_bytes % it.vec == 0
+        assert it.nbytes % it.row_bytes == 0
+        rows = it.nbytes // it.row_bytes
+        outer = 1
+        for s in it.outer_sizes:
+            outer *= s
+        assert outer == rows or (not it.outer_sizes and rows == 1)
