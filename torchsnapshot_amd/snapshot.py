@@ -564,6 +564,22 @@ class Snapshot:
         ("<rank>/<logical_path>"). ``memory_budget_bytes`` bounds peak host
         memory via tiled byte-range reads."""
         torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.read_object")
+        event_meta = {"id": uuid.uuid4().hex, "api": "read_object", "path": path}
+        log_event(Event("read_object_start", dict(event_meta)))
+        try:
+            result = self._read_object_impl(path, obj_out, memory_budget_bytes)
+            log_event(Event("read_object_end", {**event_meta, "success": True}))
+            return result
+        except Exception:
+            log_event(Event("read_object_end", {**event_meta, "success": False}))
+            raise
+
+    def _read_object_impl(
+        self,
+        path: str,
+        obj_out: Optional[Any],
+        memory_budget_bytes: Optional[int],
+    ) -> Any:
         rank_str, _, logical_path = path.partition("/")
         try:
             rank = int(rank_str)
