@@ -133,3 +133,18 @@ def test_uvm_fallbacks_on_cpu():
     t = torch.rand(4)
     assert not is_uvm_tensor(t)
     assert uvm_to_cpu(t) is t
+
+
+def test_cli_inspect_and_cat(tmp_path, capsys):
+    import torch
+
+    from torchsnapshot_amd import Snapshot, StateDict
+    from torchsnapshot_amd.__main__ import main
+
+    p = str(tmp_path / "snap")
+    Snapshot.take(p, {"sd": StateDict(w=torch.rand(16, 4), n=9)})
+    assert main(["inspect", p]) == 0
+    out = capsys.readouterr().out
+    assert "0/sd/w" in out and "tensor" in out
+    assert main(["cat", p, "0/sd/n"]) == 0
+    assert capsys.readouterr().out.strip() == "9"
