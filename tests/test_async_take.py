@@ -142,3 +142,35 @@ def test_two_concurrent_async_takes():
         s2.restore({"sd": out2})
         assert torch.equal(out1["a"], sd1["a"])
         assert torch.equal(out2["b"], sd2["b"])
+
+
+def _async_take_peer_failure(tmpdir: str) -> None:
+    """Rank 1's storage fails mid-async-drain: rank 0 must observe the
+    error through the store barrier, raise from wait(), and commit no
+    metadata."""
+    import os
+
+    import torch.distributed as dist
+
+    sd = StateDict(w=torch.rand(64, 64))
+    path = os.path.join(tmpdir, "snap")
+    if dist.get_rank() == 1:
+        with _patch_plugin(FaultyFSStoragePlugin):
+            with pytest.raises(RuntimeError):
+                pending = Snapshot.async_take(path, {"sd": sd})
+                pending.wait()
+    else:
+        pending = Snapshot.async_take(path, {"sd": sd})
+        with pytest.raises(RuntimeError):
+            pending.wait()
+    dist.barrier()
+    assert not os.path.exists(os.path.join(path, ".snapshot_metadata"))
+
+
+def test_async_take_peer_failure_world2():
+    import tempfile as tf
+
+    from torchsnapshot_amd.test_utils import run_multiprocess
+
+    with tf.TemporaryDirectory() as d:
+        run_multiprocess(2, _async_take_peer_failure, d)

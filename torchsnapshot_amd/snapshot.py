@@ -231,23 +231,41 @@ class Snapshot:
         # the commit barrier store must be created on the main thread
         # (bootstrap may use a collective)
         store = get_or_create_store(pg_wrapper)
-        pending_io_work, metadata = cls._take_impl(
-            path=path,
-            app_state=app_state,
-            storage=storage,
-            pg_wrapper=pg_wrapper,
-            replicated=replicated,
-            is_async=True,
+        barrier = LinearBarrier(
+            prefix=f"tsamd_commit_{path}",
+            store=store,
+            rank=pg_wrapper.get_rank(),
+            world_size=pg_wrapper.get_world_size(),
         )
-        # once staging is complete, the app may mutate its state freely
-        pending_io_work.wait_staged()
+        try:
+            pending_io_work, metadata = cls._take_impl(
+                path=path,
+                app_state=app_state,
+                storage=storage,
+                pg_wrapper=pg_wrapper,
+                replicated=replicated,
+                is_async=True,
+            )
+            # once staging is complete, the app may mutate its state freely
+            pending_io_work.wait_staged()
+        except Exception as e:
+            # peers are (or will be) waiting in the commit barrier: tell
+            # them this rank failed before they time out
+            try:
+                barrier.report_error(e)
+            except Exception:
+                logger.exception("failed to report early async-take error")
+            log_event(
+                Event("async_take_end", {**event_meta, "success": False})
+            )
+            raise
         return PendingSnapshot(
             path=path,
             pending_io_work=pending_io_work,
             pg_wrapper=pg_wrapper,
             metadata=metadata,
             storage=storage,
-            store=store,
+            barrier=barrier,
             storage_options=storage_options,
             event_meta=event_meta,
         )
@@ -764,7 +782,7 @@ class PendingSnapshot:
         pg_wrapper: PGWrapper,
         metadata: SnapshotMetadata,
         storage: StoragePlugin,
-        store,
+        barrier: LinearBarrier,
         storage_options: Optional[Dict[str, Any]],
         event_meta: Optional[Dict[str, Any]] = None,
     ) -> None:
@@ -775,12 +793,7 @@ class PendingSnapshot:
         self._storage = storage
         self._storage_options = storage_options
         self._event_meta = event_meta or {}
-        self._barrier = LinearBarrier(
-            prefix=f"tsamd_commit_{path}",
-            store=store,
-            rank=pg_wrapper.get_rank(),
-            world_size=pg_wrapper.get_world_size(),
-        )
+        self._barrier = barrier
         self._exc: Optional[BaseException] = None
         self._done_event = threading.Event()
         self._thread = threading.Thread(
