@@ -65,3 +65,38 @@ def test_async_take_checksums(monkeypatch):
         pending = Snapshot.async_take(path, {"sd": sd})
         pending.wait()
         assert os.path.exists(os.path.join(path, "0", ".checksums"))
+
+
+def _checksummed_dist(tmpdir: str) -> None:
+    import os
+
+    import torch.distributed as dist
+
+    os.environ["TSAMD_CHECKSUM"] = "1"
+    os.environ["TSAMD_VERIFY_CHECKSUM"] = "1"
+    torch.manual_seed(5)
+    shared = torch.rand(64, 8)
+    sd = StateDict(
+        shared=shared.clone(),
+        mine=torch.full((8,), float(dist.get_rank())),
+    )
+    path = os.path.join(tmpdir, "snap")
+    Snapshot.take(path, {"sd": sd}, replicated=["sd/shared"])
+    # each writer rank produced its checksum file
+    assert any(
+        os.path.exists(os.path.join(path, str(r), ".checksums"))
+        for r in range(dist.get_world_size())
+    )
+    out = StateDict(shared=torch.zeros(64, 8), mine=torch.zeros(8))
+    Snapshot(path).restore({"sd": out})
+    assert torch.equal(out["shared"], shared)
+    assert torch.equal(out["mine"], torch.full((8,), float(dist.get_rank())))
+
+
+def test_checksums_distributed_world2():
+    import tempfile
+
+    from torchsnapshot_amd.test_utils import run_multiprocess
+
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(2, _checksummed_dist, d)
