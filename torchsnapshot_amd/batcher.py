@@ -204,10 +204,17 @@ def _make_span(
 ) -> ReadReq:
     if len(members) == 1:
         return members[0]
+    consumer = BatchedBufferConsumer(members=members, span_start=start)
     return ReadReq(
         path=path,
         byte_range=(start, end),
-        consumer=BatchedBufferConsumer(members=members, span_start=start),
+        consumer=consumer,
+        # all-device spans read straight into pinned memory
+        buf_alloc=(
+            consumer.alloc_pinned_buffer
+            if consumer._device_fast_path_target() is not None
+            else None
+        ),
     )
 
 
@@ -215,6 +222,18 @@ class BatchedBufferConsumer(BufferConsumer):
     def __init__(self, members: List[ReadReq], span_start: int) -> None:
         self.members = members
         self.span_start = span_start
+        self._pinned_block = None
+        self._pinned_nbytes = 0
+
+    def alloc_pinned_buffer(self, nbytes: int) -> memoryview:
+        self._pinned_block = get_pinned_pool().acquire(max(nbytes, 1))
+        self._pinned_nbytes = nbytes
+        return memoryview(self._pinned_block.tensor.numpy())[:nbytes]
+
+    def close(self) -> None:
+        if self._pinned_block is not None:
+            get_pinned_pool().release(self._pinned_block)
+            self._pinned_block = None
 
     def get_consuming_cost_bytes(self) -> int:
         total = 0
@@ -256,22 +275,28 @@ class BatchedBufferConsumer(BufferConsumer):
             return
 
         def work() -> None:
-            import asyncio as _  # noqa: F401
-
             from .io_preparers.tensor import tensor_copy
-            from .ops.staging import get_pinned_pool
             from .serialization import str_to_dtype
 
-            mv = memoryview(buf)
-            nbytes = mv.nbytes
-            pool = get_pinned_pool()
-            block = pool.acquire(max(nbytes, 1))
-            try:
-                src = torch.frombuffer(mv, dtype=torch.uint8)
-                block.tensor[:nbytes].copy_(src)
-                dev_span = block.tensor[:nbytes].to(device, non_blocking=False)
-            finally:
-                pool.release(block)
+            if self._pinned_block is not None:
+                # read landed in pinned memory already: straight SDMA H2D
+                n = self._pinned_nbytes
+                dev_span = self._pinned_block.tensor[:n].to(
+                    device, non_blocking=False
+                )
+            else:
+                mv = memoryview(buf)
+                nbytes = mv.nbytes
+                pool = get_pinned_pool()
+                block = pool.acquire(max(nbytes, 1))
+                try:
+                    src = torch.frombuffer(mv, dtype=torch.uint8)
+                    block.tensor[:nbytes].copy_(src)
+                    dev_span = block.tensor[:nbytes].to(
+                        device, non_blocking=False
+                    )
+                finally:
+                    pool.release(block)
             for m in self.members:
                 c = m.consumer
                 s, e = m.byte_range
