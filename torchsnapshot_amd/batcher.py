@@ -242,25 +242,21 @@ class BatchedBufferConsumer(BufferConsumer):
         return total
 
     def _device_fast_path_target(self):
-        """If every member is a buffer-serialized tensor restore onto the
-        same CUDA device, the whole span can go up in ONE H2D and be sliced
-        on the GPU — per-member H2D round trips dominate otherwise
-        (measured 0.43 GB/s vs multi-GB/s for 2000 small tensors)."""
-        from .io_preparers.tensor import TensorBufferConsumer
-
+        """If every member consumes onto the same CUDA device, the whole
+        span goes up in ONE H2D and is sliced on the GPU — per-member H2D
+        round trips dominate otherwise (measured 0.43 GB/s vs 11.3 GB/s
+        for 2000 small tensors). Members opt in by implementing
+        device_span_target()/consume_from_device_u8() (plain tensor and
+        sharded consumers do)."""
         device = None
         for m in self.members:
-            c = m.consumer
-            if (
-                not isinstance(c, TensorBufferConsumer)
-                or c.tensor_out is None
-                or c.tensor_out.device.type != "cuda"
-                or c.entry.serializer != SERIALIZER_BUFFER
-            ):
+            getter = getattr(m.consumer, "device_span_target", None)
+            target = getter() if getter else None
+            if target is None:
                 return None
             if device is None:
-                device = c.tensor_out.device
-            elif c.tensor_out.device != device:
+                device = target
+            elif target != device:
                 return None
         return device
 
@@ -275,9 +271,6 @@ class BatchedBufferConsumer(BufferConsumer):
             return
 
         def work() -> None:
-            from .io_preparers.tensor import tensor_copy
-            from .serialization import str_to_dtype
-
             if self._pinned_block is not None:
                 # read landed in pinned memory already: straight SDMA H2D
                 n = self._pinned_nbytes
@@ -298,16 +291,8 @@ class BatchedBufferConsumer(BufferConsumer):
                 finally:
                     pool.release(block)
             for m in self.members:
-                c = m.consumer
                 s, e = m.byte_range
                 sub = dev_span[s - self.span_start : e - self.span_start]
-                dtype = str_to_dtype(c.entry.dtype)
-                loaded = (
-                    sub.view(dtype).reshape(tuple(c.entry.shape))
-                    if dtype != torch.uint8
-                    else sub.reshape(tuple(c.entry.shape))
-                )
-                tensor_copy(c.tensor_out, loaded)
-                c.fut.obj = c.tensor_out
+                m.consumer.consume_from_device_u8(sub)
 
         await asyncio.get_running_loop().run_in_executor(ctx.executor, work)

@@ -154,6 +154,26 @@ class ShardConsumer(BufferConsumer):
     def get_consuming_cost_bytes(self) -> int:
         return self.shard_entry.nbytes_estimate()
 
+    def device_span_target(self):
+        if (
+            self.all_targets_on_device()
+            and self.shard_entry.serializer != "torch_save"
+        ):
+            return self.targets[0][0].device
+        return None
+
+    def consume_from_device_u8(self, dev_u8: torch.Tensor) -> None:
+        dtype = str_to_dtype(self.shard_entry.dtype)
+        shard = (
+            dev_u8.view(dtype).reshape(tuple(self.shard_entry.shape))
+            if dtype != torch.uint8
+            else dev_u8.reshape(tuple(self.shard_entry.shape))
+        )
+        for dst_view, ov in self.targets:
+            src = narrow_nd(shard, ov.src_offsets, ov.lengths)
+            dst = narrow_nd(dst_view, ov.dst_offsets, ov.lengths)
+            tensor_copy(dst, src)
+
     async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
         def work() -> None:
             if self.shard_entry.serializer == "torch_save":

@@ -370,6 +370,27 @@ class TensorBufferConsumer(BufferConsumer):
             return 2 * nbytes
         return nbytes
 
+    def device_span_target(self):
+        # non-None when this consumer can take its bytes as a slice of a
+        # device-resident uint8 span (batched-span restore fast path)
+        if (
+            self.tensor_out is not None
+            and self.tensor_out.device.type == "cuda"
+            and self.entry.serializer == SERIALIZER_BUFFER
+        ):
+            return self.tensor_out.device
+        return None
+
+    def consume_from_device_u8(self, dev_u8: torch.Tensor) -> None:
+        dtype = str_to_dtype(self.entry.dtype)
+        loaded = (
+            dev_u8.view(dtype).reshape(tuple(self.entry.shape))
+            if dtype != torch.uint8
+            else dev_u8.reshape(tuple(self.entry.shape))
+        )
+        tensor_copy(self.tensor_out, loaded)
+        self.fut.obj = self.tensor_out
+
     async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
         def work() -> None:
             dtype = (
