@@ -111,3 +111,32 @@ def test_ddp_restore_single_process():
         sd = snapshot.get_state_dict_for_key("model")
         for k, v in net.state_dict().items():
             assert torch.equal(v, sd["module." + k])
+
+
+def _plain_save_ddp_restore(tmpdir: str) -> None:
+    """Save a PLAIN module's snapshot, restore into a DDP wrapper via
+    DDPWrappedAdapter."""
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    from torchsnapshot_amd import Snapshot
+    from torchsnapshot_amd.tricks import DDPWrappedAdapter
+
+    path = os.path.join(tmpdir, "plain")
+    if dist.get_rank() == 0:
+        torch.manual_seed(55)
+    torch.manual_seed(55)  # same weights everywhere for determinism
+    plain = _Net()
+    Snapshot.take(path, {"model": plain})
+
+    torch.manual_seed(900 + dist.get_rank())
+    ddp = DDP(_Net())
+    Snapshot(path).restore({"model": DDPWrappedAdapter(ddp)})
+    for (name, p), (_, expect) in zip(
+        ddp.module.state_dict().items(), plain.state_dict().items()
+    ):
+        assert torch.equal(p, expect), name
+
+
+def test_plain_save_restore_into_ddp():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(2, _plain_save_ddp_restore, d)

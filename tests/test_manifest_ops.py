@@ -175,3 +175,28 @@ def test_elasticity_drops_unavailable_target():
         handle_sharded_tensor_elasticity(manifest, payloads, target)
     assert "app/missing" not in target
     assert "app/k" in target
+
+
+def test_get_replicated_ranks():
+    from torchsnapshot_amd.manifest import DTensorEntry
+    from torchsnapshot_amd.manifest_utils import (
+        get_replicated_ranks,
+        is_partially_replicated_entry,
+    )
+
+    # 2x2 mesh, sharded on mesh dim 1 only -> replica sets across dim 0
+    e = DTensorEntry(
+        shards=[], mesh=[[0, 1], [2, 3]], dim_map=[[1], []],
+        dtype="float32", shape=[8, 8],
+    )
+    assert is_partially_replicated_entry(e)
+    sets = get_replicated_ranks(e)
+    assert sorted(sorted(s) for s in sets) == [[0, 2], [1, 3]]
+
+    # fully sharded: singleton sets
+    e2 = DTensorEntry(
+        shards=[], mesh=[[0, 1], [2, 3]], dim_map=[[0], [1]],
+        dtype="float32", shape=[8, 8],
+    )
+    sets = get_replicated_ranks(e2)
+    assert sorted(sorted(s) for s in sets) == [[0], [1], [2], [3]]
