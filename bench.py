@@ -143,7 +143,11 @@ def main() -> None:
 
         if use_cuda:
             torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+            dist.init_process_group(
+                backend="nccl", device_id=torch.device("cuda", local_rank)
+            )
+        else:
+            dist.init_process_group(backend="gloo")
     if use_cuda:
         device = torch.device("cuda", local_rank)
         torch.cuda.set_device(device)
@@ -166,6 +170,9 @@ def main() -> None:
     # the fs dir cache
     for _ in range(args.warmup):
         Snapshot.take(ckpt_path, app_state)
+    # drain the warmup's dirty pages so every timed step starts from the
+    # same writeback state (work outside the timed region)
+    os.sync()
     _barrier(world_size)
     if use_cuda:
         torch.cuda.synchronize(device)
