@@ -324,6 +324,9 @@ class StagedBatch:
     total_bytes: int
     _handle: Optional[int] = None          # _csnap op handle
     _device_slab: Optional[torch.Tensor] = None  # keep alive until done
+    # source tensors referenced until the async gather/copy completes, so
+    # callers may drop theirs right after stage()
+    _sources: Optional[Sequence[torch.Tensor]] = None
     _done: bool = False
 
     def wait(self) -> None:
@@ -334,6 +337,7 @@ class StagedBatch:
             _csnap.wait(self._handle)
             self._handle = None
         self._device_slab = None
+        self._sources = None
         self._done = True
 
     def memoryview_of(self, index: int) -> memoryview:
@@ -399,6 +403,7 @@ class StagingEngine:
         if total == 0:
             batch._done = True
             return batch
+        batch._sources = tensors
         try:
             if self._use_ext:
                 self._stage_ext(tensors, items, batch, total)
