@@ -234,9 +234,16 @@ void chain_after(hipStream_t side, uintptr_t producer_stream, int device) {
   // stream (the training step that materialized the tensors)
   hipEvent_t dep;
   HIP_CHECK(hipEventCreateWithFlags(&dep, hipEventDisableTiming));
-  HIP_CHECK(hipEventRecord(dep, reinterpret_cast<hipStream_t>(producer_stream)));
-  HIP_CHECK(hipStreamWaitEvent(side, dep, 0));
-  HIP_CHECK(hipEventDestroy(dep));
+  hipError_t e =
+      hipEventRecord(dep, reinterpret_cast<hipStream_t>(producer_stream));
+  if (e == hipSuccess) {
+    e = hipStreamWaitEvent(side, dep, 0);
+  }
+  (void)hipEventDestroy(dep);  // stream holds its own reference
+  if (e != hipSuccess) {
+    throw std::runtime_error(std::string("HIP error chaining streams: ") +
+                             hipGetErrorString(e));
+  }
 }
 
 // flat desc rows from python: 18 u64 each (see ops/staging.py)
@@ -425,19 +432,23 @@ static void op_wait(long long handle) {
     ev = it->second.ev;
     device = it->second.device;
   }
+  hipError_t e;
   {
     py::gil_scoped_release release;
-    hipError_t e = hipEventSynchronize(ev);
-    if (e != hipSuccess) {
-      throw std::runtime_error(std::string("hipEventSynchronize: ") +
-                               hipGetErrorString(e));
+    e = hipEventSynchronize(ev);
+  }
+  {
+    // erase even on failure so the op map and event never leak
+    std::lock_guard<std::mutex> lk(g_mu);
+    auto it = g_ops.find(handle);
+    if (it != g_ops.end()) {
+      (void)hipEventDestroy(it->second.ev);
+      g_ops.erase(it);
     }
   }
-  std::lock_guard<std::mutex> lk(g_mu);
-  auto it = g_ops.find(handle);
-  if (it != g_ops.end()) {
-    hipEventDestroy(it->second.ev);
-    g_ops.erase(it);
+  if (e != hipSuccess) {
+    throw std::runtime_error(std::string("hipEventSynchronize: ") +
+                             hipGetErrorString(e));
   }
 }
 
