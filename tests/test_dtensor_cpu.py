@@ -155,3 +155,36 @@ def test_dtensor_read_object_full():
 
         out = Snapshot(os.path.join(d, "snap")).read_object("0/obj/dt")
         assert torch.equal(out, _full())
+
+
+def _save_replicated_with_glob(tmpdir: str) -> None:
+    """A fully-replicated DTensor under replicated=['**'] must not lose
+    its payload to partitioner reassignment (the preparer already chose
+    its writer)."""
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import distribute_tensor
+    from torch.distributed.tensor.placement_types import Replicate
+
+    from torchsnapshot_amd import Snapshot
+    from torchsnapshot_amd.state_dict import StateDict
+
+    mesh = init_device_mesh("cpu", (dist.get_world_size(),))
+    dt = distribute_tensor(_full(), mesh, [Replicate()])
+    torch.manual_seed(11)
+    plain = torch.rand(32, 4)
+    sd = StateDict(dt=dt, plain=plain)
+    path = os.path.join(tmpdir, "snap")
+    Snapshot.take(path, {"app": sd}, replicated=["**"])
+
+    out = StateDict(
+        dt=distribute_tensor(torch.zeros(48, 8), mesh, [Replicate()]),
+        plain=torch.zeros(32, 4),
+    )
+    Snapshot(path).restore({"app": out})
+    assert torch.equal(out["dt"].full_tensor(), _full())
+    assert torch.equal(out["plain"], plain)
+
+
+def test_replicated_dtensor_with_glob_world2():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(2, _save_replicated_with_glob, d)

@@ -49,7 +49,6 @@ from .manifest_ops import (
     get_manifest_for_rank,
     handle_sharded_tensor_elasticity,
 )
-from .manifest_utils import is_fully_replicated_entry
 from .partitioner import PartitionItem, partition_write_reqs
 from .pg_wrapper import PGWrapper
 from .rng_state import RNGState
@@ -361,8 +360,13 @@ class Snapshot:
         non_replicated_bytes = 0
         for req in write_reqs:
             logical = req_to_logical.get(req.path)
-            if logical in replicated_paths and is_fully_replicated_entry(
-                manifest[logical]
+            # only plain replicated-marked entries participate: DTensor /
+            # ShardedTensor writes are already deduplicated by their
+            # preparers (replica-set round robin), and reassigning them
+            # here would drop payloads whose chosen writer has the only
+            # write request
+            if logical in replicated_paths and getattr(
+                manifest[logical], "replicated", False
             ):
                 items.append(
                     PartitionItem(
