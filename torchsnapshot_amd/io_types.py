@@ -116,6 +116,11 @@ class StoragePlugin(abc.ABC):
     async def close(self) -> None:
         ...
 
+    async def close_for_loop(self) -> None:
+        """Release resources tied to the CURRENT event loop (called when a
+        pipeline's loop ends). Backends holding per-loop sessions override
+        this; default no-op."""
+
     # -- sync conveniences (run on a private event loop) --------------------
 
     def sync_write(self, write_io: WriteIO) -> None:

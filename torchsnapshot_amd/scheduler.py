@@ -294,6 +294,12 @@ def execute_write_reqs(
             signaler.cancel()
             reporter.cancel()
             executor.shutdown(wait=False)
+            try:
+                # release per-event-loop resources (network backends keep
+                # one session per loop; this loop dies with the pipeline)
+                await storage.close_for_loop()
+            except Exception:
+                logger.exception("storage close at pipeline end failed")
 
     return _spawn_pipeline(main, stats, staged_event, done_event, checksums)
 
@@ -397,6 +403,10 @@ def execute_read_reqs(
         finally:
             reporter.cancel()
             executor.shutdown(wait=False)
+            try:
+                await storage.close_for_loop()
+            except Exception:
+                logger.exception("storage close at pipeline end failed")
 
     return _spawn_pipeline(main, stats, staged_event, done_event)
 

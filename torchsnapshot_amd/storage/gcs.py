@@ -221,7 +221,10 @@ class GCSStoragePlugin(StoragePlugin):
 
         run_coro_sync(go())
 
-    async def close(self) -> None:
+    async def close_for_loop(self) -> None:
         sess = self._sessions.pop(id(asyncio.get_running_loop()), None)
         if sess is not None and not sess.closed:
             await sess.close()
+
+    async def close(self) -> None:
+        await self.close_for_loop()
