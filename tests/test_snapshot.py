@@ -250,3 +250,24 @@ def test_snapshot_delete_fs():
         assert not os.path.exists(path)
         with pytest.raises(RuntimeError):
             _ = Snapshot(path).metadata
+
+
+def test_non_dict_state():
+    """A stateful whose state_dict() returns a bare tensor (not a dict)."""
+
+    class Weird:
+        def __init__(self, t):
+            self.t = t
+
+        def state_dict(self):
+            return self.t
+
+        def load_state_dict(self, sd):
+            self.t = sd
+
+    w = Weird(torch.rand(7, 3))
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"w": w})
+        w2 = Weird(torch.zeros(7, 3))
+        snap.restore({"w": w2})
+        assert torch.equal(w2.t, w.t)
