@@ -36,6 +36,7 @@ def main() -> None:
     parser.add_argument("--work-dir", default="/tmp/tsamd_ddp_bench")
     parser.add_argument("--device", default="cuda")
     parser.add_argument("--compare-torch-save", action="store_true")
+    parser.add_argument("--benchmark-load", action="store_true")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -76,6 +77,18 @@ def main() -> None:
             f"({total_bytes / 1e9 / elapsed:.2f} GB/s aggregate, "
             f"{world_size} ranks)"
         )
+
+    if args.benchmark_load:
+        t0 = time.monotonic()
+        Snapshot(path).restore({"model": sd})
+        if world_size > 1:
+            dist.barrier()
+        elapsed = time.monotonic() - t0
+        if rank == 0:
+            print(
+                f"torchsnapshot_amd restore: {elapsed:.2f}s "
+                f"({total_bytes / 1e9 / elapsed:.2f} GB/s aggregate)"
+            )
 
     if args.compare_torch_save and rank == 0:
         # free the snapshot's disk space first: the torch.save copy of a

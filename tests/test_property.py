@@ -90,3 +90,66 @@ def test_collapse_layout_any_view(dims, data):
         for s in it.outer_sizes:
             outer *= s
         assert outer == rows or (not it.outer_sizes and rows == 1)
+
+
+@settings(max_examples=60, deadline=None)
+@given(data=st.data())
+def test_manifest_entry_json_round_trip(data):
+    import json
+
+    from torchsnapshot_amd.manifest import (
+        ChunkedTensorEntry,
+        PrimitiveEntry,
+        Shard,
+        TensorEntry,
+        entry_from_dict,
+    )
+
+    kind = data.draw(st.sampled_from(["tensor", "chunked", "primitive"]))
+    if kind == "tensor":
+        entry = TensorEntry(
+            location=data.draw(st.text(max_size=12)),
+            serializer=data.draw(st.sampled_from(["buffer", "torch_save", "qtensor"])),
+            dtype="float32",
+            shape=data.draw(st.lists(st.integers(0, 64), max_size=4)),
+            replicated=data.draw(st.booleans()),
+            byte_range=data.draw(
+                st.one_of(st.none(), st.tuples(st.integers(0, 100), st.integers(100, 200)).map(list))
+            ),
+        )
+    elif kind == "chunked":
+        n = data.draw(st.integers(1, 3))
+        entry = ChunkedTensorEntry(
+            dtype="bfloat16",
+            shape=[n * 4, 2],
+            chunks=[
+                Shard(
+                    offsets=[i * 4, 0],
+                    sizes=[4, 2],
+                    tensor=TensorEntry(
+                        location=f"l{i}", serializer="buffer",
+                        dtype="bfloat16", shape=[4, 2],
+                    ),
+                )
+                for i in range(n)
+            ],
+            replicated=data.draw(st.booleans()),
+        )
+    else:
+        value = data.draw(
+            st.one_of(
+                st.integers(-(2**50), 2**50),
+                st.floats(allow_nan=False),
+                st.booleans(),
+                st.text(max_size=12),
+                st.binary(max_size=12),
+            )
+        )
+        entry = PrimitiveEntry.from_object(value)
+        rebuilt = entry_from_dict(json.loads(json.dumps(entry.to_dict())))
+        out = rebuilt.get_value()
+        assert type(out) is type(value) and out == value
+        return
+
+    rebuilt = entry_from_dict(json.loads(json.dumps(entry.to_dict())))
+    assert rebuilt.to_dict() == entry.to_dict()
