@@ -271,3 +271,19 @@ def test_non_dict_state():
         w2 = Weird(torch.zeros(7, 3))
         snap.restore({"w": w2})
         assert torch.equal(w2.t, w.t)
+
+
+def test_snapshot_relocatable():
+    """All payload locations are relative: a moved snapshot directory
+    restores identically."""
+    import shutil
+
+    sd = StateDict(a=torch.rand(32, 8), n=5)
+    with tmp_snapshot_path() as path:
+        Snapshot.take(path, {"sd": sd})
+        moved = path + "_moved"
+        shutil.move(path, moved)
+        out = StateDict()
+        Snapshot(moved).restore({"sd": out})
+        assert torch.equal(out["a"], sd["a"])
+        assert out["n"] == 5
