@@ -129,6 +129,19 @@ class Snapshot:
                 self._metadata = SnapshotMetadata.from_str(
                     bytes(read_io.buf).decode("utf-8")
                 )
+                from .version import __version__ as _v
+
+                if (
+                    self._metadata.version.split(".")[0]
+                    != _v.split(".")[0]
+                ):
+                    logger.warning(
+                        "snapshot at %s was written by version %s; this "
+                        "build is %s — load will be attempted anyway",
+                        self.path,
+                        self._metadata.version,
+                        _v,
+                    )
             finally:
                 storage.sync_close()
         return self._metadata
