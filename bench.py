@@ -197,6 +197,15 @@ def main() -> None:
     pending.wait()
     stall_s = _max_over_ranks(stall_s, world_size, device)
 
+    # informational: one restore of the full state (untimed region for the
+    # headline metric; reported as restore_GBps)
+    _barrier(world_size)
+    t0 = time.monotonic()
+    Snapshot(ckpt_path).restore({"model": state})
+    restore_s = _max_over_ranks(
+        time.monotonic() - t0, world_size, device
+    )
+
     if rank == 0 and not args.keep:
         shutil.rmtree(ckpt_path, ignore_errors=True)
 
@@ -216,6 +225,7 @@ def main() -> None:
             "dtype": "bf16",
             "data": "synthetic",
             "stall_sec": round(stall_s, 3),
+            "restore_GBps": round((total_bytes / 1e9) / restore_s, 3),
             "config": {
                 "model": "llama-3-8b" if args.model != "tiny" else "tiny",
                 "model_bytes": total_bytes,
