@@ -253,3 +253,32 @@ def test_stage_flipped_tensor():
     batch.wait()
     assert bytes(batch.memoryview_of(0)) == _ref_bytes(f)
     batch.release()
+
+
+def test_gpu_saved_cpu_restored():
+    """Cross-device: snapshot taken from device tensors restores into CPU
+    tensors (and vice versa)."""
+    sd = StateDict(
+        big=torch.randn(1024, 1024, dtype=torch.bfloat16, device="cuda"),
+        small1=torch.randn(100, device="cuda"),
+        small2=torch.randn(64, 64, device="cuda"),
+    )
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict(
+            big=torch.zeros(1024, 1024, dtype=torch.bfloat16),
+            small1=torch.zeros(100),
+            small2=torch.zeros(64, 64),
+        )
+        snap.restore({"sd": out})
+        assert torch.equal(out["big"], sd["big"].cpu())
+        assert torch.equal(out["small1"], sd["small1"].cpu())
+        assert torch.equal(out["small2"], sd["small2"].cpu())
+
+    # CPU-saved -> GPU-restored
+    cpu_sd = StateDict(w=torch.rand(256, 128))
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": cpu_sd})
+        out = StateDict(w=torch.zeros(256, 128, device="cuda"))
+        snap.restore({"sd": out})
+        assert torch.equal(out["w"].cpu(), cpu_sd["w"])
