@@ -121,3 +121,26 @@ def _save_chunked_replicated(tmpdir: str) -> None:
 def test_chunked_replicated_spread():
     with tempfile.TemporaryDirectory() as d:
         run_multiprocess(2, _save_chunked_replicated, d)
+
+
+def _save_partitioner_disabled(tmpdir: str) -> None:
+    """TSAMD_DISABLE_PARTITIONER: every rank writes replicated payloads
+    (identical bytes to identical paths — safe); restore still works."""
+    import torchsnapshot_amd.knobs as knobs
+    from torchsnapshot_amd import Snapshot
+    from torchsnapshot_amd.state_dict import StateDict
+
+    torch.manual_seed(0)
+    shared = torch.rand(64, 16)
+    sd = StateDict(shared=shared.clone())
+    path = os.path.join(tmpdir, "snap")
+    with knobs.override_env("TSAMD_DISABLE_PARTITIONER", "1"):
+        Snapshot.take(path, {"app": sd}, replicated=["**"])
+    out = StateDict(shared=torch.zeros(64, 16))
+    Snapshot(path).restore({"app": out})
+    assert torch.equal(out["shared"], shared)
+
+
+def test_partitioner_disabled_world2():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(2, _save_partitioner_disabled, d)
