@@ -287,3 +287,22 @@ def test_snapshot_relocatable():
         Snapshot(moved).restore({"sd": out})
         assert torch.equal(out["a"], sd["a"])
         assert out["n"] == 5
+
+
+def test_corrupted_metadata_message():
+    with tmp_snapshot_path() as path:
+        Snapshot.take(path, {"sd": StateDict(a=1)})
+        with open(os.path.join(path, ".snapshot_metadata"), "w") as f:
+            f.write("{not valid json or yaml: [")
+        with pytest.raises(ValueError, match="corrupted"):
+            _ = Snapshot(path).metadata
+
+
+def test_fsync_mode(monkeypatch):
+    monkeypatch.setenv("TSAMD_FSYNC", "1")
+    sd = StateDict(a=torch.rand(64, 16))
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict()
+        snap.restore({"sd": out})
+        assert torch.equal(out["a"], sd["a"])

@@ -408,9 +408,22 @@ class SnapshotMetadata:
                 loader = yaml.CSafeLoader
             except AttributeError:
                 loader = yaml.SafeLoader
-            d = yaml.load(s, Loader=loader)
-        return cls(
-            version=d["version"],
-            world_size=int(d["world_size"]),
-            manifest={k: entry_from_dict(v) for k, v in d["manifest"].items()},
-        )
+            try:
+                d = yaml.load(s, Loader=loader)
+            except yaml.YAMLError as e:
+                raise ValueError(
+                    "snapshot metadata is corrupted (neither valid JSON "
+                    f"nor YAML): {e}"
+                ) from e
+        try:
+            return cls(
+                version=d["version"],
+                world_size=int(d["world_size"]),
+                manifest={
+                    k: entry_from_dict(v) for k, v in d["manifest"].items()
+                },
+            )
+        except (KeyError, TypeError, AttributeError) as e:
+            raise ValueError(
+                f"snapshot metadata is corrupted (missing/invalid field: {e})"
+            ) from e

@@ -21,6 +21,14 @@ from .. import knobs
 from ..io_types import ReadIO, StoragePlugin, WriteIO
 
 
+def _fsync_enabled() -> bool:
+    """TSAMD_FSYNC=1: fsync every payload (and its directory) so a
+    committed snapshot survives power loss, not just process crashes.
+    Costs raw-disk write throughput; page-cache-speed saves are the
+    default (the reference's aiofiles plugin never syncs either)."""
+    return os.environ.get("TSAMD_FSYNC", "0") not in ("0", "", "false")
+
+
 class FSStoragePlugin(StoragePlugin):
     def __init__(self, root: str, storage_options: Optional[dict] = None) -> None:
         self.root = root
@@ -55,8 +63,18 @@ class FSStoragePlugin(StoragePlugin):
             total = len(mv)
             while off < total:
                 off += os.pwrite(fd, mv[off : off + chunk], off)
+            if _fsync_enabled():
+                os.fsync(fd)
         finally:
             os.close(fd)
+        if _fsync_enabled():
+            # make the directory entry durable too (crash consistency: the
+            # metadata commit must never be durable before its payloads)
+            dfd = os.open(os.path.dirname(full), os.O_RDONLY)
+            try:
+                os.fsync(dfd)
+            finally:
+                os.close(dfd)
 
     def _read_sync(self, read_io: ReadIO) -> None:
         full = self._abspath(read_io.path)
