@@ -21,10 +21,13 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 def test_sigkill_mid_save_leaves_no_commit():
     with tempfile.TemporaryDirectory() as d:
         path = os.path.join(d, "snap")
+        env = dict(os.environ, PYTHONPATH=REPO)
         proc = subprocess.Popen(
             [sys.executable, os.path.join(REPO, "tests", "_crash_child.py"), path],
             cwd=REPO,
+            env=env,
             stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE,
             text=True,
         )
         # wait until the child is inside take(), then kill it hard
