@@ -306,3 +306,19 @@ def test_fsync_mode(monkeypatch):
         out = StateDict()
         snap.restore({"sd": out})
         assert torch.equal(out["a"], sd["a"])
+
+
+def test_custom_tensor_prepare_func():
+    """Save-time transform: persist fp32 weights as bf16."""
+    sd = StateDict(w=torch.rand(64, 32))
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(
+            path,
+            {"sd": sd},
+            _custom_tensor_prepare_func=lambda p, t: t.to(torch.bfloat16),
+        )
+        assert snap.get_manifest()["0/sd/w"]["dtype"] == "bfloat16"
+        out = StateDict(w=torch.zeros(64, 32))
+        snap.restore({"sd": out})
+        assert out["w"].dtype == torch.float32  # cast back on load
+        assert torch.equal(out["w"], sd["w"].to(torch.bfloat16).float())

@@ -179,6 +179,7 @@ class Snapshot:
         pg: Optional[dist.ProcessGroup] = None,
         replicated: Optional[List[str]] = None,
         storage_options: Optional[Dict[str, Any]] = None,
+        _custom_tensor_prepare_func: Optional[Any] = None,
     ) -> "Snapshot":
         torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.take")
         cls._validate_app_state(app_state)
@@ -199,6 +200,7 @@ class Snapshot:
                     pg_wrapper=pg_wrapper,
                     replicated=replicated,
                     is_async=False,
+                    custom_tensor_prepare_func=_custom_tensor_prepare_func,
                 )
                 pending_io_work.complete()
                 from .integrity import write_checksum_file
@@ -291,6 +293,7 @@ class Snapshot:
         pg_wrapper: PGWrapper,
         replicated: List[str],
         is_async: bool,
+        custom_tensor_prepare_func: Optional[Any] = None,
     ) -> Tuple[PendingIOWork, SnapshotMetadata]:
         rank = pg_wrapper.get_rank()
         world_size = pg_wrapper.get_world_size()
@@ -321,6 +324,15 @@ class Snapshot:
         replicated_paths = cls._calculate_replicated_entries(
             flattened, replicated, pg_wrapper
         )
+
+        if custom_tensor_prepare_func is not None:
+            # save-time tensor transform hook (e.g. cast fp32 weights to
+            # bf16 in the snapshot); mirrors the reference's private
+            # _custom_tensor_prepare_func (snapshot.py take/_take_impl)
+            for p in list(flattened.keys()):
+                obj = flattened[p]
+                if isinstance(obj, torch.Tensor):
+                    flattened[p] = custom_tensor_prepare_func(p, obj)
 
         write_reqs: List[WriteReq] = []
         req_to_logical: Dict[str, str] = {}
