@@ -322,3 +322,61 @@ def test_custom_tensor_prepare_func():
         snap.restore({"sd": out})
         assert out["w"].dtype == torch.float32  # cast back on load
         assert torch.equal(out["w"], sd["w"].to(torch.bfloat16).float())
+
+
+def test_tied_weights_written_once():
+    """Tied parameters (same tensor under two state-dict keys) produce ONE
+    payload; both paths restore."""
+
+    class Tied(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.embed = torch.nn.Embedding(64, 16)
+            self.head = torch.nn.Linear(16, 64, bias=False)
+            self.head.weight = self.embed.weight  # tie
+
+    m = Tied()
+    with tmp_snapshot_path() as path:
+        with __import__("torchsnapshot_amd").knobs.override_batching_disabled(True):
+            snap = Snapshot.take(path, {"m": m})
+        manifest = snap.get_manifest()
+        e1 = manifest["0/m/embed.weight"]
+        e2 = manifest["0/m/head.weight"]
+        assert e1["location"] == e2["location"]
+        # exactly one payload file for the tied weight
+        payloads = []
+        for root, _, names in os.walk(path):
+            payloads += [os.path.join(root, n) for n in names if "weight" in n]
+        assert len(payloads) == 1, payloads
+
+        m2 = Tied()
+        snap.restore({"m": m2})
+        assert torch.equal(m2.embed.weight, m.embed.weight)
+        assert torch.equal(m2.head.weight, m.head.weight)
+        assert m2.head.weight.data_ptr() == m2.embed.weight.data_ptr()
+
+
+def test_tied_weights_batched():
+    class Tied(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = torch.nn.Linear(32, 32, bias=False)
+            self.b = torch.nn.Linear(32, 32, bias=False)
+            self.c = torch.nn.Linear(32, 32, bias=False)
+            self.b.weight = self.a.weight
+
+    m = Tied()
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"m": m})
+        manifest = snap.get_manifest()
+        assert (
+            manifest["0/m/a.weight"]["location"]
+            == manifest["0/m/b.weight"]["location"]
+        )
+        assert manifest["0/m/a.weight"].get("byte_range") == manifest[
+            "0/m/b.weight"
+        ].get("byte_range")
+        m2 = Tied()
+        snap.restore({"m": m2})
+        assert torch.equal(m2.a.weight, m.a.weight)
+        assert torch.equal(m2.c.weight, m.c.weight)

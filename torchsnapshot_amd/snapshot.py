@@ -336,7 +336,29 @@ class Snapshot:
 
         write_reqs: List[WriteReq] = []
         req_to_logical: Dict[str, str] = {}
+        # tied-weight dedup: identical tensor objects (same storage, view,
+        # dtype — e.g. tied input/output embeddings) are written once; the
+        # aliases SHARE the writer's entry object, so batcher relocation
+        # stays consistent for every path
+        seen_tensors: Dict[Any, Entry] = {}
         for logical_path, obj in flattened.items():
+            alias_key = None
+            if (
+                isinstance(obj, torch.Tensor)
+                and type(obj) is torch.Tensor
+                and not obj.is_quantized
+            ):
+                alias_key = (
+                    obj.data_ptr(),
+                    obj.dtype,
+                    tuple(obj.shape),
+                    tuple(obj.stride()),
+                    str(obj.device),
+                    logical_path in replicated_paths,
+                )
+                if alias_key in seen_tensors:
+                    manifest[logical_path] = seen_tensors[alias_key]
+                    continue
             entry, reqs = prepare_write(
                 obj=obj,
                 logical_path=logical_path,
@@ -345,6 +367,8 @@ class Snapshot:
                 is_async_snapshot=is_async,
             )
             manifest[logical_path] = entry
+            if alias_key is not None:
+                seen_tensors[alias_key] = entry
             for r in reqs:
                 req_to_logical[r.path] = logical_path
             write_reqs.extend(reqs)
