@@ -282,3 +282,24 @@ def test_gpu_saved_cpu_restored():
         out = StateDict(w=torch.zeros(256, 128, device="cuda"))
         snap.restore({"sd": out})
         assert torch.equal(out["w"].cpu(), cpu_sd["w"])
+
+
+def test_tied_weights_on_gpu():
+    class Tied(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.embed = torch.nn.Embedding(128, 64)
+            self.head = torch.nn.Linear(64, 128, bias=False)
+            self.head.weight = self.embed.weight
+
+    m = Tied().cuda()
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"m": m})
+        man = snap.get_manifest()
+        assert (
+            man["0/m/embed.weight"]["location"]
+            == man["0/m/head.weight"]["location"]
+        )
+        m2 = Tied().cuda()
+        snap.restore({"m": m2})
+        assert torch.equal(m2.embed.weight, m.embed.weight)
