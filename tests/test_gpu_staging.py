@@ -303,3 +303,28 @@ def test_tied_weights_on_gpu():
         m2 = Tied().cuda()
         snap.restore({"m": m2})
         assert torch.equal(m2.embed.weight, m.embed.weight)
+
+
+def test_async_stall_much_smaller_than_sync():
+    """The headline property: async_take returns (staging done) in a small
+    fraction of the full sync save time."""
+    import time
+
+    sd = StateDict(
+        **{
+            f"w{i}": torch.randn(16, 1024, 1024, dtype=torch.bfloat16, device="cuda")
+            for i in range(32)
+        }
+    )  # 1 GB
+    with tmp_snapshot_path() as path:
+        t0 = time.monotonic()
+        Snapshot.take(path, {"sd": sd})
+        sync_s = time.monotonic() - t0
+    with tmp_snapshot_path() as path:
+        t0 = time.monotonic()
+        pending = Snapshot.async_take(path, {"sd": sd})
+        stall_s = time.monotonic() - t0
+        pending.wait()
+    # staging 1 GB at ~50 GB/s is ~0.02 s; storage is the long pole.
+    # generous bound to stay robust on shared boxes:
+    assert stall_s < max(0.5 * sync_s, 0.5), (stall_s, sync_s)
