@@ -328,3 +328,55 @@ def test_async_stall_much_smaller_than_sync():
     # staging 1 GB at ~50 GB/s is ~0.02 s; storage is the long pole.
     # generous bound to stay robust on shared boxes:
     assert stall_s < max(0.5 * sync_s, 0.5), (stall_s, sync_s)
+
+
+def test_device_psum64_matches_cpu(monkeypatch):
+    """The gather kernel's psum64 must match the CPU verifier exactly, for
+    every layout class and both stage modes."""
+    from torchsnapshot_amd import integrity
+
+    torch.manual_seed(3)
+    tensors = [
+        torch.randn(333, 257, device="cuda"),                      # contig
+        torch.randn(256, 256, device="cuda").t(),                  # transpose
+        torch.randn(100, 64, device="cuda")[:, ::2],               # strided
+        torch.randn(4096, 64, dtype=torch.bfloat16, device="cuda"),# bf16
+        torch.randn(17, device="cuda"),                            # tail < 8B mult
+        (torch.rand(77, 3, device="cuda") * 255).to(torch.uint8),  # 1-byte
+    ]
+    for mode in ("direct", "slab"):
+        monkeypatch.setenv("TSAMD_STAGE_MODE", mode)
+        engine = staging.get_staging_engine(tensors[0].device)
+        batch = engine.stage(tensors, compute_checksums=True)
+        batch.wait()
+        assert batch.checksums is not None
+        total = 0
+        for i, t in enumerate(tensors):
+            mv = batch.memoryview_of(i)
+            # per-tensor checksum uses FILE offsets; recompute via shifting:
+            # the CPU check below covers the whole-file identity instead
+            total = (total + batch.checksums[i]) % (1 << 64)
+        want = integrity.psum64_hexdigest(batch.slab_memoryview())
+        got = "psum64:" + format(total, "016x")
+        assert got == want, (mode, got, want)
+        batch.release()
+
+
+def test_checksummed_gpu_snapshot_verifies(monkeypatch):
+    monkeypatch.setenv("TSAMD_CHECKSUM", "1")
+    monkeypatch.setenv("TSAMD_VERIFY_CHECKSUM", "1")
+    sd = StateDict(
+        big=torch.randn(2048, 2048, dtype=torch.bfloat16, device="cuda"),
+        small=torch.randn(100, device="cuda"),
+        tr=torch.randn(128, 128, device="cuda").t(),
+    )
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict(
+            big=torch.zeros(2048, 2048, dtype=torch.bfloat16, device="cuda"),
+            small=torch.zeros(100, device="cuda"),
+            tr=torch.zeros(128, 128, device="cuda"),
+        )
+        snap.restore({"sd": out})
+        assert torch.equal(out["big"], sd["big"])
+        assert torch.equal(out["tr"], sd["tr"].contiguous())

@@ -120,6 +120,7 @@ class BatchedBufferStager(BufferStager):
         self.is_async_snapshot = is_async_snapshot
         self._staged_batch = None
         self._pinned_block = None
+        self.precomputed_checksum = None
 
     def get_staging_cost_bytes(self) -> int:
         return self.total_bytes
@@ -132,9 +133,17 @@ class BatchedBufferStager(BufferStager):
         return await loop.run_in_executor(ctx.executor, self._stage_cpu)
 
     def _stage_device(self) -> BufferType:
+        from . import integrity
+
         engine = get_staging_engine(self.tensors[0].device)
-        batch = engine.stage(self.tensors)
+        ck = integrity.checksumming_enabled()
+        batch = engine.stage(self.tensors, compute_checksums=ck)
         batch.wait()
+        if ck and batch.checksums is not None:
+            # padding is zeroed, so the slab-file checksum is the sum of
+            # the per-member device checksums
+            total = sum(batch.checksums) % (1 << 64)
+            self.precomputed_checksum = "psum64:" + format(total, "016x")
         self._staged_batch = batch
         return batch.slab_memoryview()
 

@@ -35,7 +35,52 @@ def verification_enabled() -> bool:
 def hash_buffer(buf) -> str:
     import xxhash
 
-    return xxhash.xxh3_64_hexdigest(buf)
+    return "xxh3:" + xxhash.xxh3_64_hexdigest(buf)
+
+
+# -- psum64: the device-side checksum (computed for free inside the gather
+# kernel; see ops/hip/staging.hip). An order-independent weighted word sum:
+# sum(word_i * (splitmix64(i) | 1)) mod 2^64 over little-endian u64 words,
+# zero-padded tail. Verified here vectorized with numpy. ---------------------
+
+
+def _splitmix64_mult(idx):
+    import numpy as np
+
+    z = idx + np.uint64(0x9E3779B97F4A7C15)
+    z = (z ^ (z >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+    z = (z ^ (z >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+    z = z ^ (z >> np.uint64(31))
+    return z | np.uint64(1)
+
+
+def psum64_hexdigest(buf) -> str:
+    import numpy as np
+
+    mv = memoryview(buf)
+    if mv.format != "B":
+        mv = mv.cast("B")
+    total = np.uint64(0)
+    chunk_words = 4 * 1024 * 1024  # 32 MB pieces keep index arrays small
+    nbytes = mv.nbytes
+    word_base = 0
+    off = 0
+    old = np.seterr(over="ignore")
+    try:
+        while off < nbytes:
+            end = min(off + chunk_words * 8, nbytes)
+            piece = np.frombuffer(mv[off:end], dtype=np.uint8)
+            pad = (-len(piece)) % 8
+            if pad:
+                piece = np.concatenate([piece, np.zeros(pad, dtype=np.uint8)])
+            words = piece.view("<u8")
+            idx = np.arange(word_base, word_base + len(words), dtype=np.uint64)
+            total = total + np.uint64((words * _splitmix64_mult(idx)).sum())
+            word_base += len(words)
+            off = end
+    finally:
+        np.seterr(**old)
+    return "psum64:" + format(int(total), "016x")
 
 
 def checksum_file_path(rank: int) -> str:
@@ -77,7 +122,12 @@ def verify_buffer(path: str, buf, expected: Dict[str, str]) -> None:
     want = expected.get(path)
     if want is None:
         return
-    got = hash_buffer(buf)
+    if want.startswith("psum64:"):
+        got = psum64_hexdigest(buf)
+    else:
+        got = hash_buffer(buf)
+        if not want.startswith("xxh3:"):
+            got = got.removeprefix("xxh3:")  # legacy untagged
     if got != want:
         raise RuntimeError(
             f"checksum mismatch for payload '{path}': snapshot recorded "

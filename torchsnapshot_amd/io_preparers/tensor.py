@@ -267,6 +267,8 @@ class TensorBufferStager(BufferStager):
         self.serializer = serializer
         self.is_async_snapshot = is_async_snapshot
         self._staged_batch = None  # StagedBatch for device tensors
+        # device staging computes this for free when TSAMD_CHECKSUM=1
+        self.precomputed_checksum = None
 
     def get_staging_cost_bytes(self) -> int:
         nbytes = self.tensor.numel() * self.tensor.element_size()
@@ -299,9 +301,16 @@ class TensorBufferStager(BufferStager):
             if self.is_async_snapshot:
                 cpu = cpu.clone()
             return tensor_as_memoryview(cpu.contiguous())
+        from .. import integrity
+
         engine = get_staging_engine(t.device)
-        batch = engine.stage([t])
+        ck = integrity.checksumming_enabled()
+        batch = engine.stage([t], compute_checksums=ck)
         batch.wait()  # blocks in executor thread; GIL released inside HIP
+        if ck and batch.checksums is not None:
+            self.precomputed_checksum = "psum64:" + format(
+                batch.checksums[0], "016x"
+            )
         self._staged_batch = batch
         return batch.memoryview_of(0)
 

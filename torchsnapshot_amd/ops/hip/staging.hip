@@ -66,6 +66,48 @@ struct Desc {
   long long strides[kMaxDims];   // byte strides of outer dims
 };
 
+// ---------------------------------------------------------------------------
+// psum64 checksum: an order-independent weighted word sum over the flat
+// payload. contribution(byte b, value v) treats the payload as little-endian
+// u64 words at FILE offsets: word w = flat_offset+b >> 3, lane = b & 7, and
+// adds (v << 8*lane) * M(w) where M(w) = splitmix64(w) | 1. Linear in the
+// value bytes, so any write width (16/8/4/2/1) contributes independently —
+// the CPU verifier just does sum(words * M(index)) over the whole file
+// (padding bytes are zeroed, contributing nothing).
+// ---------------------------------------------------------------------------
+
+__device__ __host__ inline unsigned long long psum_mult(unsigned long long w) {
+  // splitmix64 finalizer
+  unsigned long long z = w + 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z = z ^ (z >> 31);
+  return z | 1ull;  // odd multiplier
+}
+
+template <int VEC>
+__device__ inline unsigned long long psum_contrib(
+    const char* bytes, unsigned long long file_off) {
+  unsigned long long acc = 0;
+  if (VEC == 16) {
+    const unsigned long long* p = reinterpret_cast<const unsigned long long*>(bytes);
+    unsigned long long w = file_off >> 3;
+    acc += p[0] * psum_mult(w);
+    acc += p[1] * psum_mult(w + 1);
+  } else if (VEC == 8) {
+    acc += *reinterpret_cast<const unsigned long long*>(bytes) *
+           psum_mult(file_off >> 3);
+  } else {
+    // narrow writes: shift the piece into its word lane
+    unsigned long long v = 0;
+    if (VEC == 4) v = *reinterpret_cast<const uint32_t*>(bytes);
+    else if (VEC == 2) v = *reinterpret_cast<const uint16_t*>(bytes);
+    else v = *reinterpret_cast<const unsigned char*>(bytes);
+    acc += (v << (8 * (file_off & 7))) * psum_mult(file_off >> 3);
+  }
+  return acc;
+}
+
 // byte offset of row `row` within the strided tensor (excluding columns)
 __device__ inline unsigned long long row_offset(const Desc& d, uint32_t row) {
   unsigned long long off = 0;
@@ -108,20 +150,23 @@ __device__ inline void copy_vec(char* flat, const char* strided) {
   }
 }
 
-// Copy [s, e) of the tensor's flat byte range for one work unit.
-template <int VEC, bool GATHER>
-__device__ void process_range(const Desc& d, char* flat_base, uint32_t s,
-                              uint32_t e) {
+// Copy [s, e) of the tensor's flat byte range for one work unit. When
+// HASH (gather only), returns this thread's psum64 contribution.
+template <int VEC, bool GATHER, bool HASH>
+__device__ unsigned long long process_range(const Desc& d, char* flat_base,
+                                            uint32_t s, uint32_t e) {
   char* flat = flat_base + d.flat_off;
   const uint32_t tid = threadIdx.x;
   const uint32_t rb = d.row_bytes;
+  unsigned long long h = 0;
   if (d.ndim == 0 || rb == d.nbytes) {
     // fully contiguous: plain vectorized copy of [s, e)
     const char* src = d.tensor_base;
     for (uint32_t i = s + tid * VEC; i < e; i += kBlockThreads * VEC) {
       copy_vec<VEC, GATHER>(flat + i, src + i);
+      if (HASH) h += psum_contrib<VEC>(src + i, d.flat_off + i);
     }
-    return;
+    return h;
   }
   if (rb >= kWideRowBytes) {
     // wide rows: rows sequential, lanes across columns
@@ -133,6 +178,7 @@ __device__ void process_range(const Desc& d, char* flat_base, uint32_t s,
       uint32_t n = min(rb - col, e - off);
       for (uint32_t i = tid * VEC; i < n; i += kBlockThreads * VEC) {
         copy_vec<VEC, GATHER>(flat + off + i, src + col + i);
+        if (HASH) h += psum_contrib<VEC>(src + col + i, d.flat_off + off + i);
       }
       off += n;
       ++row;
@@ -149,16 +195,19 @@ __device__ void process_range(const Desc& d, char* flat_base, uint32_t s,
       uint32_t hi = (flat_pos + rb > e) ? (e - flat_pos) : rb;
       for (uint32_t i = lo; i < hi; i += VEC) {
         copy_vec<VEC, GATHER>(flat + flat_pos + i, src + i);
+        if (HASH) h += psum_contrib<VEC>(src + i, d.flat_off + flat_pos + i);
       }
     }
   }
+  return h;
 }
 
-template <bool GATHER>
+template <bool GATHER, bool HASH>
 __global__ __launch_bounds__(kBlockThreads) void pack_kernel(
     const Desc* __restrict__ descs, int n,
     const unsigned long long* __restrict__ wu_prefix,
-    unsigned long long total_wus, char* __restrict__ flat_base) {
+    unsigned long long total_wus, char* __restrict__ flat_base,
+    unsigned long long* __restrict__ hash_out) {
   for (unsigned long long wu = blockIdx.x; wu < total_wus; wu += gridDim.x) {
     // binary search: which tensor owns this work unit
     int lo = 0, hi = n - 1;
@@ -172,12 +221,22 @@ __global__ __launch_bounds__(kBlockThreads) void pack_kernel(
     uint32_t s = (uint32_t)(local_wu * kWorkUnitBytes);
     uint32_t e = s + kWorkUnitBytes > d.nbytes ? d.nbytes
                                                : s + kWorkUnitBytes;
+    unsigned long long h = 0;
     switch (d.vec) {
-      case 16: process_range<16, GATHER>(d, flat_base, s, e); break;
-      case 8: process_range<8, GATHER>(d, flat_base, s, e); break;
-      case 4: process_range<4, GATHER>(d, flat_base, s, e); break;
-      case 2: process_range<2, GATHER>(d, flat_base, s, e); break;
-      default: process_range<1, GATHER>(d, flat_base, s, e); break;
+      case 16: h = process_range<16, GATHER, HASH>(d, flat_base, s, e); break;
+      case 8: h = process_range<8, GATHER, HASH>(d, flat_base, s, e); break;
+      case 4: h = process_range<4, GATHER, HASH>(d, flat_base, s, e); break;
+      case 2: h = process_range<2, GATHER, HASH>(d, flat_base, s, e); break;
+      default: h = process_range<1, GATHER, HASH>(d, flat_base, s, e); break;
+    }
+    if (HASH) {
+      // wave64 reduction, then one atomic per wave per work unit
+      for (int delta = 32; delta > 0; delta >>= 1) {
+        h += __shfl_down(h, delta, 64);
+      }
+      if ((threadIdx.x & 63) == 0 && h != 0) {
+        atomicAdd(&hash_out[lo], h);
+      }
     }
   }
 }
@@ -282,7 +341,7 @@ std::vector<Desc> parse_descs(const std::vector<unsigned long long>& flat,
 long long launch_pack(const std::vector<unsigned long long>& flat, int n,
                       uintptr_t slab_ptr, uintptr_t pinned_ptr,
                       unsigned long long total_bytes, uintptr_t producer_stream,
-                      int device, bool gather) {
+                      int device, bool gather, uintptr_t hash_out_ptr) {
   HIP_CHECK(hipSetDevice(device));
   DeviceCtx& ctx = get_ctx(device);
   hipStream_t stream = gather ? ctx.d2h_stream : ctx.h2d_stream;
@@ -347,12 +406,20 @@ long long launch_pack(const std::vector<unsigned long long>& flat, int n,
   }
   unsigned int grid = (unsigned int)std::min<unsigned long long>(
       total_wus == 0 ? 1 : total_wus, grid_cap);
-  if (gather) {
-    hipLaunchKernelGGL(pack_kernel<true>, dim3(grid), dim3(kBlockThreads), 0,
-                       stream, d_descs, n, d_prefix, total_wus, flat_base);
+  unsigned long long* hash_out =
+      reinterpret_cast<unsigned long long*>(hash_out_ptr);
+  if (gather && hash_out != nullptr) {
+    hipLaunchKernelGGL((pack_kernel<true, true>), dim3(grid),
+                       dim3(kBlockThreads), 0, stream, d_descs, n, d_prefix,
+                       total_wus, flat_base, hash_out);
+  } else if (gather) {
+    hipLaunchKernelGGL((pack_kernel<true, false>), dim3(grid),
+                       dim3(kBlockThreads), 0, stream, d_descs, n, d_prefix,
+                       total_wus, flat_base, nullptr);
   } else {
-    hipLaunchKernelGGL(pack_kernel<false>, dim3(grid), dim3(kBlockThreads), 0,
-                       stream, d_descs, n, d_prefix, total_wus, flat_base);
+    hipLaunchKernelGGL((pack_kernel<false, false>), dim3(grid),
+                       dim3(kBlockThreads), 0, stream, d_descs, n, d_prefix,
+                       total_wus, flat_base, nullptr);
   }
   HIP_CHECK(hipGetLastError());
 
@@ -409,9 +476,10 @@ static long long h2d_copy(uintptr_t src_pinned, uintptr_t dst,
 static long long pack_d2h(const std::vector<unsigned long long>& flat, int n,
                           uintptr_t slab_ptr, uintptr_t pinned_ptr,
                           unsigned long long total_bytes,
-                          uintptr_t producer_stream, int device) {
+                          uintptr_t producer_stream, int device,
+                          uintptr_t hash_out_ptr = 0) {
   return launch_pack(flat, n, slab_ptr, pinned_ptr, total_bytes,
-                     producer_stream, device, /*gather=*/true);
+                     producer_stream, device, /*gather=*/true, hash_out_ptr);
 }
 
 static long long scatter_h2d(const std::vector<unsigned long long>& flat, int n,
@@ -419,7 +487,7 @@ static long long scatter_h2d(const std::vector<unsigned long long>& flat, int n,
                              unsigned long long total_bytes,
                              uintptr_t producer_stream, int device) {
   return launch_pack(flat, n, slab_ptr, pinned_ptr, total_bytes,
-                     producer_stream, device, /*gather=*/false);
+                     producer_stream, device, /*gather=*/false, 0);
 }
 
 static void op_wait(long long handle) {
@@ -514,7 +582,11 @@ PYBIND11_MODULE(_csnap, m) {
   m.def("d2h_copy", &d2h_copy, "async D2H copy on the side stream");
   m.def("h2d_copy", &h2d_copy, "async H2D copy on the side stream");
   m.def("pack_d2h", &pack_d2h,
-        "gather-pack tensors into a slab and copy it to pinned host memory");
+        "gather-pack tensors into a slab and copy it to pinned host memory",
+        py::arg("flat"), py::arg("n"), py::arg("slab_ptr"),
+        py::arg("pinned_ptr"), py::arg("total_bytes"),
+        py::arg("producer_stream"), py::arg("device"),
+        py::arg("hash_out_ptr") = 0);
   m.def("scatter_h2d", &scatter_h2d,
         "copy pinned host bytes to device and scatter into strided tensors");
   m.def("wait", &op_wait, "block until an op completes");

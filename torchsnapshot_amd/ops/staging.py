@@ -324,6 +324,10 @@ class StagedBatch:
     total_bytes: int
     _handle: Optional[int] = None          # _csnap op handle
     _device_slab: Optional[torch.Tensor] = None  # keep alive until done
+    # per-tensor psum64 values (device-computed), populated at wait() when
+    # checksumming was requested
+    checksums: Optional[List[int]] = None
+    _hash_tensor: Optional[torch.Tensor] = None
     # source tensors referenced until the async gather/copy completes, so
     # callers may drop theirs right after stage()
     _sources: Optional[Sequence[torch.Tensor]] = None
@@ -336,6 +340,9 @@ class StagedBatch:
         if self._handle is not None:
             _csnap.wait(self._handle)
             self._handle = None
+        if self._hash_tensor is not None:
+            self.checksums = [int(v) for v in self._hash_tensor.cpu().tolist()]
+            self._hash_tensor = None
         self._device_slab = None
         self._sources = None
         self._done = True
@@ -369,8 +376,17 @@ class StagingEngine:
         # before the first checkpoint
         warm_pinned_pool()
 
-    def stage(self, tensors: Sequence[torch.Tensor]) -> StagedBatch:
+    def stage(
+        self,
+        tensors: Sequence[torch.Tensor],
+        compute_checksums: bool = False,
+    ) -> StagedBatch:
         """Start async D2H staging of device tensors into one pinned slab.
+
+        With ``compute_checksums``, the gather kernel also accumulates a
+        psum64 checksum per tensor at no measurable cost (the kernel is
+        PCIe-bound); padding gaps are zeroed so the sum of the per-tensor
+        values is the checksum of the whole written file.
 
         The caller must ensure the tensors' producing stream is
         torch.cuda.current_stream() of this thread (true for checkpointing:
@@ -406,7 +422,10 @@ class StagingEngine:
         batch._sources = tensors
         try:
             if self._use_ext:
-                self._stage_ext(tensors, items, batch, total)
+                self._stage_ext(
+                    tensors, items, batch, total,
+                    compute_checksums=compute_checksums,
+                )
             else:
                 self._stage_torch_fallback(tensors, items, batch)
         except BaseException:
@@ -423,11 +442,16 @@ class StagingEngine:
         items: List[PackItem],
         batch: StagedBatch,
         total: int,
+        compute_checksums: bool = False,
     ) -> None:
         dev_index = self.device.index or 0
         with torch.cuda.device(dev_index):
             cur_stream = torch.cuda.current_stream().cuda_stream
-            single_contig = len(items) == 1 and not items[0].outer_sizes
+            single_contig = (
+                len(items) == 1
+                and not items[0].outer_sizes
+                and not compute_checksums
+            )
             if single_contig:
                 # one SDMA copy, no kernel
                 handle = _csnap.d2h_copy(
@@ -441,11 +465,34 @@ class StagingEngine:
                 mode = _pack_mode()
                 slab = None
                 slab_ptr = 0
+                hash_ptr = 0
+                if compute_checksums:
+                    # padding must read as zeros so per-tensor checksums
+                    # sum to the file checksum
+                    batch._hash_tensor = torch.zeros(
+                        len(items), dtype=torch.uint64, device=self.device
+                    )
+                    hash_ptr = batch._hash_tensor.data_ptr()
                 if mode == "slab":
-                    slab = torch.empty(
-                        total, dtype=torch.uint8, device=self.device
+                    slab = (
+                        torch.zeros(total, dtype=torch.uint8, device=self.device)
+                        if compute_checksums
+                        else torch.empty(
+                            total, dtype=torch.uint8, device=self.device
+                        )
                     )
                     slab_ptr = slab.data_ptr()
+                elif compute_checksums:
+                    # direct mode: kernel writes only payload bytes; zero
+                    # the pinned gap regions host-side (tiny)
+                    pin_np = batch.pinned.tensor.numpy()
+                    prev_end = 0
+                    for it in items:
+                        if it.flat_offset > prev_end:
+                            pin_np[prev_end : it.flat_offset] = 0
+                        prev_end = it.flat_offset + it.nbytes
+                    if total > prev_end:
+                        pin_np[prev_end:total] = 0
                 handle = _csnap.pack_d2h(
                     _items_to_flat(items),
                     len(items),
@@ -454,6 +501,7 @@ class StagingEngine:
                     total,
                     cur_stream,
                     dev_index,
+                    hash_ptr,
                 )
                 batch._device_slab = slab
             batch._handle = handle

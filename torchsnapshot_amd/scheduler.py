@@ -253,8 +253,12 @@ def execute_write_reqs(
                     stats.stage_s += time.monotonic() - t0
                 nbytes = memoryview(buf).nbytes
                 if do_checksum:
-                    checksums[req.path] = await asyncio.get_running_loop(
-                    ).run_in_executor(executor, integrity.hash_buffer, buf)
+                    pre = getattr(req.stager, "precomputed_checksum", None)
+                    if pre is not None:
+                        checksums[req.path] = pre
+                    else:
+                        checksums[req.path] = await asyncio.get_running_loop(
+                        ).run_in_executor(executor, integrity.hash_buffer, buf)
                 stats.staged_reqs += 1
                 stats.staged_bytes += nbytes
                 staged_remaining -= 1
