@@ -121,6 +121,9 @@ class BatchedBufferStager(BufferStager):
         self._staged_batch = None
         self._pinned_block = None
         self.precomputed_checksum = None
+        # [(start, end, "psum64:<hex>")] per member, recorded so restore
+        # can verify byte-range/merged-span reads of this slab
+        self.member_checksums = None
 
     def get_staging_cost_bytes(self) -> int:
         return self.total_bytes
@@ -144,21 +147,45 @@ class BatchedBufferStager(BufferStager):
             # the per-member device checksums
             total = sum(batch.checksums) % (1 << 64)
             self.precomputed_checksum = "psum64:" + format(total, "016x")
+            self.member_checksums = [
+                (off, off + n, "psum64:" + format(v % (1 << 64), "016x"))
+                for off, n, v in zip(
+                    batch.offsets, batch.nbytes_list, batch.checksums
+                )
+            ]
         self._staged_batch = batch
         return batch.slab_memoryview()
 
     def _stage_cpu(self) -> BufferType:
+        from . import integrity
+
         items, offsets, total = build_pack_items(self.tensors)
         slab = bytearray(total)
         mv = memoryview(slab)
         from .serialization import tensor_as_memoryview
 
+        ck = integrity.checksumming_enabled()
+        members = []
         for t, off in zip(self.tensors, offsets):
             nbytes = t.numel() * t.element_size()
             if nbytes == 0:
                 continue
             src = tensor_as_memoryview(t if t.is_contiguous() else t.contiguous())
             mv[off : off + nbytes] = src
+            if ck:
+                members.append(
+                    (
+                        off,
+                        off + nbytes,
+                        integrity.psum64_hexdigest(src, word_base=off // 8),
+                    )
+                )
+        if ck:
+            self.member_checksums = members
+            total_ck = sum(
+                int(v[len("psum64:"):], 16) for _, _, v in members
+            ) % (1 << 64)
+            self.precomputed_checksum = "psum64:" + format(total_ck, "016x")
         return mv
 
     def release_buffer(self) -> None:

@@ -54,7 +54,11 @@ def _splitmix64_mult(idx):
     return z | np.uint64(1)
 
 
-def psum64_hexdigest(buf) -> str:
+def psum64_hexdigest(buf, word_base: int = 0) -> str:
+    """psum64 of ``buf``. ``word_base`` is the u64 word index of the
+    buffer's first byte within the containing file, so subrange checksums
+    of byte-range reads line up with the file-global indexing used by the
+    device kernel (ops/hip/staging.hip)."""
     import numpy as np
 
     mv = memoryview(buf)
@@ -63,7 +67,7 @@ def psum64_hexdigest(buf) -> str:
     total = np.uint64(0)
     chunk_words = 4 * 1024 * 1024  # 32 MB pieces keep index arrays small
     nbytes = mv.nbytes
-    word_base = 0
+    word_base = np.uint64(word_base)
     off = 0
     old = np.seterr(over="ignore")
     try:
@@ -116,6 +120,55 @@ def load_checksums(
         found = True
         merged.update(json.loads(bytes(read_io.buf).decode("utf-8")))
     return merged if found else None
+
+
+def member_key(path: str, start: int, end: int) -> str:
+    """Checksum-file key for one batched-slab member's byte range."""
+    return f"{path}#{start}-{end}"
+
+
+def verify_ranged_buffer(
+    path: str, buf, byte_range, expected: Dict[str, str]
+) -> None:
+    """Verify a byte-range read (a batched-slab member or a merged span of
+    members) against per-member psum64 values recorded at save time.
+
+    psum64 is additive over disjoint file word-ranges and slab padding is
+    zeroed, so the expected checksum of any member-aligned span is the
+    (mod 2^64) sum of the recorded per-member values inside it. Ranges not
+    covered by any recorded member (e.g. tiled reads of an unbatched
+    tensor, or snapshots saved without checksumming) are skipped.
+    """
+    start, end = byte_range
+    prefix = path + "#"
+    total = 0
+    found = False
+    for k, v in expected.items():
+        if not k.startswith(prefix) or not v.startswith("psum64:"):
+            continue
+        s_str, _, e_str = k[len(prefix):].partition("-")
+        s, e = int(s_str), int(e_str)
+        if s >= start and e <= end:
+            total = (total + int(v[len("psum64:"):], 16)) % (1 << 64)
+            found = True
+    if not found:
+        return
+    if start % 8 != 0:
+        # slab members are 64-byte aligned; a misaligned span can't use
+        # file-global word indexing — don't verify rather than misreport
+        logger.warning(
+            "skipping checksum verification of misaligned span %s[%d:%d]",
+            path, start, end,
+        )
+        return
+    got = psum64_hexdigest(buf, word_base=start // 8)
+    want = "psum64:" + format(total, "016x")
+    if got != want:
+        raise RuntimeError(
+            f"checksum mismatch for span '{path}' [{start}:{end}): snapshot "
+            f"recorded {want}, read back {got} — the file is corrupted or "
+            "was modified after the snapshot was committed"
+        )
 
 
 def verify_buffer(path: str, buf, expected: Dict[str, str]) -> None:

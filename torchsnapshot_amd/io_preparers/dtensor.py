@@ -110,6 +110,16 @@ class DTensorIOPreparer:
         write_reqs: List[WriteReq] = []
         pieces = subdivide_shard(local, global_offset, max_bytes)
         for i, (piece, piece_offsets) in enumerate(pieces):
+            # round-robin writer within the replica set. ONLY the writer
+            # emits the shard entry: non-writer copies would keep the
+            # standalone location after the writer's batcher relocates the
+            # payload into a slab, and the load-side merge could pick the
+            # stale copy (round-1 advisor finding). The load-side merge
+            # unions shards across all ranks, so readers still see the
+            # full shard set.
+            writer = replica_set[i % len(replica_set)]
+            if writer != my_rank:
+                continue
             location = location_for_shard(storage_path, piece_offsets)
             sub_entry, sub_reqs = TensorIOPreparer.prepare_write(
                 storage_path=location,
@@ -124,10 +134,7 @@ class DTensorIOPreparer:
                     tensor=sub_entry,
                 )
             )
-            # round-robin writer within the replica set
-            writer = replica_set[i % len(replica_set)]
-            if writer == my_rank:
-                write_reqs.extend(sub_reqs)
+            write_reqs.extend(sub_reqs)
         entry = DTensorEntry(
             shards=shards_meta,
             mesh=obj.device_mesh.mesh.tolist(),

@@ -59,7 +59,11 @@ def get_manifest_for_rank(
 
     for key, entry in metadata.manifest.items():
         w_rank, path = _split_rank_path(key)
-        if is_sharded_entry(entry):
+        if is_sharded_entry(entry) or isinstance(entry, DTensorEntry):
+            # every DTensor entry goes through the shard-union merge, even
+            # fully-replicated ones: each replica-set rank only records the
+            # shard pieces it wrote (round-robin), so no single rank's
+            # entry holds the full set
             sharded_by_path[path].append((w_rank, entry))
         elif is_fully_replicated_entry(entry) and not is_container_entry(entry):
             replicated_by_path[path].append((w_rank, entry))

@@ -259,6 +259,12 @@ def execute_write_reqs(
                     else:
                         checksums[req.path] = await asyncio.get_running_loop(
                         ).run_in_executor(executor, integrity.hash_buffer, buf)
+                    # per-member values so byte-range/merged-span reads of
+                    # batched slabs are verifiable on restore
+                    for s, e, v in (
+                        getattr(req.stager, "member_checksums", None) or []
+                    ):
+                        checksums[integrity.member_key(req.path, s, e)] = v
                 stats.staged_reqs += 1
                 stats.staged_bytes += nbytes
                 staged_remaining -= 1
@@ -384,11 +390,17 @@ def execute_read_reqs(
                     stats.io_s += time.monotonic() - t0
                 buf = read_io.buf
                 stats.io_bytes += memoryview(buf).nbytes
-                if checksums and req.byte_range is None:
-                    await asyncio.get_running_loop().run_in_executor(
-                        executor, integrity.verify_buffer, req.path, buf,
-                        checksums,
-                    )
+                if checksums:
+                    if req.byte_range is None:
+                        await asyncio.get_running_loop().run_in_executor(
+                            executor, integrity.verify_buffer, req.path, buf,
+                            checksums,
+                        )
+                    else:
+                        await asyncio.get_running_loop().run_in_executor(
+                            executor, integrity.verify_ranged_buffer,
+                            req.path, buf, req.byte_range, checksums,
+                        )
                 t0 = time.monotonic()
                 await req.consumer.consume_buffer(ctx, buf)
                 stats.consume_s += time.monotonic() - t0

@@ -188,7 +188,7 @@ class Snapshot:
         event_meta = {"id": unique_id, "rank": pg_wrapper.get_rank(), "api": "take"}
         log_event(Event("take_start", dict(event_meta)))
         try:
-            path, replicated = cls._coalesce_path_and_replicated(
+            path, replicated, _ = cls._coalesce_path_and_replicated(
                 path, app_state, replicated or [], pg_wrapper
             )
             storage = url_to_storage_plugin(path, storage_options)
@@ -238,15 +238,17 @@ class Snapshot:
             "api": "async_take",
         }
         log_event(Event("async_take_start", dict(event_meta)))
-        path, replicated = cls._coalesce_path_and_replicated(
+        path, replicated, snapshot_uid = cls._coalesce_path_and_replicated(
             path, app_state, replicated or [], pg_wrapper
         )
         storage = url_to_storage_plugin(path, storage_options)
         # the commit barrier store must be created on the main thread
-        # (bootstrap may use a collective)
+        # (bootstrap may use a collective); its key namespace is unique per
+        # snapshot so repeated saves to the same path (or a save after a
+        # failed one) never see stale counters or a stale error_flag
         store = get_or_create_store(pg_wrapper)
         barrier = LinearBarrier(
-            prefix=f"tsamd_commit_{path}",
+            prefix=f"tsamd_commit_{snapshot_uid}",
             store=store,
             rank=pg_wrapper.get_rank(),
             world_size=pg_wrapper.get_world_size(),
@@ -438,21 +440,28 @@ class Snapshot:
                 logical = req_to_logical[req.path]
                 dropped_logical.setdefault(logical, set()).add(req.path)
 
-        # a rank's manifest only claims replicated payloads it writes
+        # a rank's manifest only claims replicated payloads it writes.
+        # Tied-weight aliases SHARE the dropped entry object under other
+        # logical paths — drop/replace every path holding the identical
+        # object, or stale standalone locations (which the writer's batcher
+        # relocates away from) would survive in this rank's manifest.
         for logical, dropped in dropped_logical.items():
             entry = manifest[logical]
+            alias_paths = [p for p, e in manifest.items() if e is entry]
             if isinstance(entry, ChunkedTensorEntry):
                 kept_chunks = [
                     c for c in entry.chunks if c.tensor.location not in dropped
                 ]
                 if kept_chunks:
-                    manifest[logical] = dataclass_replace(
-                        entry, chunks=kept_chunks
-                    )
+                    new_entry = dataclass_replace(entry, chunks=kept_chunks)
+                    for p in alias_paths:
+                        manifest[p] = new_entry
                 else:
-                    del manifest[logical]
+                    for p in alias_paths:
+                        del manifest[p]
             else:
-                del manifest[logical]
+                for p in alias_paths:
+                    del manifest[p]
         return kept_reqs
 
     @classmethod
@@ -706,9 +715,12 @@ class Snapshot:
         app_state: AppState,
         replicated: List[str],
         pg_wrapper: PGWrapper,
-    ) -> Tuple[str, List[str]]:
-        # all ranks must agree on the snapshot path: rank 0 wins
-        obj_list = [path]
+    ) -> Tuple[str, List[str], str]:
+        # all ranks must agree on the snapshot path: rank 0 wins. A fresh
+        # uid rides along so the async-commit barrier gets a unique store
+        # namespace per snapshot (stale arrive/depart counters or a prior
+        # failure's error_flag under the same path must not leak in).
+        obj_list: List[Any] = [path, uuid.uuid4().hex]
         pg_wrapper.broadcast_object_list(obj_list, src=0)
         if obj_list[0] != path:
             logger.warning(
@@ -718,7 +730,7 @@ class Snapshot:
                 path,
                 obj_list[0],
             )
-        path = obj_list[0]
+        path, uid = obj_list[0], obj_list[1]
 
         replicated = list(replicated) + cls._infer_replicated(app_state)
         # a pattern counts only if every rank requested it
@@ -734,7 +746,7 @@ class Snapshot:
                 "ignoring them",
                 sorted(dropped),
             )
-        return path, sorted(common)
+        return path, sorted(common), uid
 
     @staticmethod
     def _infer_replicated(app_state: AppState) -> List[str]:
