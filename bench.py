@@ -97,12 +97,29 @@ def build_state(
         from torch.distributed.tensor.placement_types import Shard
 
         mesh = init_device_mesh(device.type, (world_size,))
+        rank = int(os.environ.get("RANK", "0"))
         for name, shape in shapes:
-            assert shape[0] % world_size == 0, (name, shape)
-            local_shape = (shape[0] // world_size,) + tuple(shape[1:])
+            # uneven Shard(0) split with torch.chunk semantics (what
+            # DTensor uses): ceil-sized chunks, trailing ranks may hold
+            # fewer (or zero) rows — no divisibility requirement
+            chunk = -(-shape[0] // world_size)
+            lo = min(rank * chunk, shape[0])
+            hi = min(lo + chunk, shape[0])
+            local_shape = (hi - lo,) + tuple(shape[1:])
             local = torch.empty(local_shape, dtype=dtype, device=device)
-            local = local.float().normal_(0, 0.02).to(dtype)
-            sd[name] = DTensor.from_local(local, mesh, [Shard(0)])
+            if local.numel():
+                local = local.float().normal_(0, 0.02).to(dtype)
+            full_stride = torch.empty(
+                shape, dtype=dtype, device="meta"
+            ).stride()
+            sd[name] = DTensor.from_local(
+                local,
+                mesh,
+                [Shard(0)],
+                run_check=False,
+                shape=torch.Size(shape),
+                stride=full_stride,
+            )
             total_bytes += int(
                 torch.Size(shape).numel() * local.element_size()
             )

@@ -82,3 +82,40 @@ def test_bench_tiny_world2_gloo():
         assert out["n_gpus"] == 2
         assert out["config"]["parallelism"] == "fsdp2"
         assert out["ms_per_step"] > 0
+
+
+def test_bench_tiny_world3_uneven_gloo():
+    """World 3 makes every tiny shape split unevenly (64 -> 22/22/20,
+    16 -> 6/6/4): the bench (and the DTensor save path under it) must not
+    require divisible shards."""
+    with tempfile.TemporaryDirectory() as d:
+        r = _run(
+            [
+                sys.executable,
+                "-m",
+                "torch.distributed.run",
+                "--nnodes=1",
+                "--nproc-per-node",
+                "3",
+                "--master-addr",
+                "127.0.0.1",
+                "--master-port",
+                "29513",
+                "bench.py",
+                "--device",
+                "cpu",
+                "--model",
+                "tiny",
+                "--steps",
+                "1",
+                "--warmup",
+                "1",
+                "--dir",
+                d,
+            ]
+        )
+        assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+        lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+        out = json.loads(lines[-1])
+        assert out["n_gpus"] == 3
+        assert out["ms_per_step"] > 0

@@ -188,3 +188,45 @@ def _save_replicated_with_glob(tmpdir: str) -> None:
 def test_replicated_dtensor_with_glob_world2():
     with tempfile.TemporaryDirectory() as d:
         run_multiprocess(2, _save_replicated_with_glob, d)
+
+
+# ---------------------------------------------------------------------------
+# uneven (indivisible) DTensor shards: world 3 save, world 5 restore
+# ---------------------------------------------------------------------------
+
+
+def _uneven_full(seed: int = 21) -> torch.Tensor:
+    torch.manual_seed(seed)
+    return torch.rand(47, 8)  # 47 rows: uneven at world 3 and 5
+
+
+def _save_uneven(tmpdir: str) -> None:
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import distribute_tensor
+    from torch.distributed.tensor.placement_types import Shard
+
+    from torchsnapshot_amd import Snapshot
+
+    mesh = init_device_mesh("cpu", (dist.get_world_size(),))
+    dt = distribute_tensor(_uneven_full(), mesh, [Shard(0)])
+    Snapshot.take(os.path.join(tmpdir, "snap"), {"obj": _Holder(dt)})
+
+
+def _restore_uneven(tmpdir: str) -> None:
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import distribute_tensor
+    from torch.distributed.tensor.placement_types import Shard
+
+    from torchsnapshot_amd import Snapshot
+
+    mesh = init_device_mesh("cpu", (dist.get_world_size(),))
+    dt = distribute_tensor(torch.zeros(47, 8), mesh, [Shard(0)])
+    holder = _Holder(dt)
+    Snapshot(os.path.join(tmpdir, "snap")).restore({"obj": holder})
+    assert torch.equal(holder.dt.full_tensor(), _uneven_full())
+
+
+def test_dtensor_uneven_save3_restore5():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(3, _save_uneven, d)
+        run_multiprocess(5, _restore_uneven, d)

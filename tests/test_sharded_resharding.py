@@ -132,3 +132,116 @@ def test_load_into_full_tensor():
     out = torch.zeros(10, 4)
     _simulate(shards, [(out, [0, 0])])
     assert torch.equal(out, full)
+
+
+def _row_views(out: torch.Tensor, parts: int):
+    """torch.chunk-style (ceil, possibly-empty-tail) row split of ``out``."""
+    n = out.shape[0]
+    step = -(-n // parts)
+    views = []
+    for i in range(parts):
+        lo = min(i * step, n)
+        hi = min(lo + step, n)
+        if hi > lo:
+            views.append((out[lo:hi], [lo] + [0] * (out.dim() - 1)))
+    return views
+
+
+@pytest.mark.parametrize("dim", [0, 1])
+@pytest.mark.parametrize("src_parts", [1, 2, 3, 5])
+@pytest.mark.parametrize("dst_parts", [1, 2, 4, 7])
+def test_reshard_matrix_dims_counts(dim, src_parts, dst_parts):
+    """The reference's ChunkShardingSpec matrix: every (shard dim, source
+    count, dest count) combination, sizes chosen indivisible (47, 23)
+    (reference tests/test_sharded_tensor_resharding.py:98-110)."""
+    full = torch.rand(47, 23)
+    n = full.shape[dim]
+
+    def split(parts):
+        step = -(-n // parts)
+        pieces = []
+        for i in range(parts):
+            lo = min(i * step, n)
+            hi = min(lo + step, n)
+            if hi > lo:
+                off = [0, 0]
+                off[dim] = lo
+                pieces.append((off, full.narrow(dim, lo, hi - lo).clone()))
+        return pieces
+
+    shards = split(src_parts)
+    out = torch.zeros(47, 23)
+    views = []
+    step = -(-n // dst_parts)
+    for i in range(dst_parts):
+        lo = min(i * step, n)
+        hi = min(lo + step, n)
+        if hi > lo:
+            off = [0, 0]
+            off[dim] = lo
+            views.append((out.narrow(dim, lo, hi - lo), off))
+    _simulate(shards, views)
+    assert torch.equal(out, full)
+
+
+def test_reshard_enumerable_irregular():
+    """EnumerableShardingSpec-style irregular tiling: explicitly-placed
+    unequal rectangles covering the tensor, restored into row-wise views
+    and into a 2-D grid."""
+    full = torch.rand(10, 12)
+    shards = [
+        ([0, 0], full[0:3, 0:12].clone()),          # wide top strip
+        ([3, 0], full[3:10, 0:5].clone()),          # tall left block
+        ([3, 5], full[3:6, 5:12].clone()),          # mid right block
+        ([6, 5], full[6:10, 5:9].clone()),          # lower middle
+        ([6, 9], full[6:10, 9:12].clone()),         # lower right
+    ]
+    out = torch.zeros(10, 12)
+    _simulate(shards, _row_views(out, 4))
+    assert torch.equal(out, full)
+
+    out2 = torch.zeros(10, 12)
+    grid_views = [
+        (out2[0:5, 0:6], [0, 0]),
+        (out2[0:5, 6:12], [0, 6]),
+        (out2[5:10, 0:6], [5, 0]),
+        (out2[5:10, 6:12], [5, 6]),
+    ]
+    _simulate(shards, grid_views)
+    assert torch.equal(out2, full)
+
+
+def test_reshard_random_tilings():
+    """Randomized rectangular tilings on both sides (seeded): recursively
+    split the tensor into irregular tiles, persist one tiling, restore
+    into another."""
+    g = torch.Generator().manual_seed(1234)
+
+    def tile(lo0, hi0, lo1, hi1, depth):
+        if depth == 0 or (hi0 - lo0 < 2 and hi1 - lo1 < 2):
+            return [(lo0, hi0, lo1, hi1)]
+        if (hi0 - lo0 >= 2) and (
+            hi1 - lo1 < 2 or int(torch.randint(0, 2, (1,), generator=g)) == 0
+        ):
+            cut = lo0 + 1 + int(
+                torch.randint(0, hi0 - lo0 - 1, (1,), generator=g)
+            )
+            return tile(lo0, cut, lo1, hi1, depth - 1) + tile(
+                cut, hi0, lo1, hi1, depth - 1
+            )
+        cut = lo1 + 1 + int(torch.randint(0, hi1 - lo1 - 1, (1,), generator=g))
+        return tile(lo0, hi0, lo1, cut, depth - 1) + tile(
+            lo0, hi0, cut, hi1, depth - 1
+        )
+
+    for trial in range(4):
+        full = torch.rand(21, 17)
+        src = tile(0, 21, 0, 17, 3)
+        dst = tile(0, 21, 0, 17, 3)
+        shards = [
+            ([a, c], full[a:b, c:d].clone()) for a, b, c, d in src
+        ]
+        out = torch.zeros(21, 17)
+        views = [(out[a:b, c:d], [a, c]) for a, b, c, d in dst]
+        _simulate(shards, views)
+        assert torch.equal(out, full), f"trial {trial}"

@@ -95,3 +95,67 @@ def test_sharded_read_into_full_tensor():
 
         out = Snapshot(os.path.join(d, "snap")).read_object("0/obj/st")
         assert torch.equal(out, _full_reference())
+
+
+# ---------------------------------------------------------------------------
+# indivisible table, world 3 -> world 5 (uneven shards on both sides)
+# ---------------------------------------------------------------------------
+
+_IND_ROWS = 47  # indivisible by 3 and by 5
+
+
+def _make_indivisible(seed: int):
+    from torch.distributed._shard import sharded_tensor
+    from torch.distributed._shard.sharding_spec import ChunkShardingSpec
+
+    world_size = dist.get_world_size()
+    spec = ChunkShardingSpec(
+        dim=0,
+        placements=[f"rank:{r}/cpu" for r in range(world_size)],
+    )
+    st = sharded_tensor.rand(spec, (_IND_ROWS, 16))
+    full = _indivisible_reference(seed)
+    for shard in st.local_shards():
+        lo = shard.metadata.shard_offsets[0]
+        rows = shard.tensor.shape[0]
+        shard.tensor.copy_(full[lo : lo + rows])
+    return st
+
+
+def _indivisible_reference(seed: int) -> torch.Tensor:
+    torch.manual_seed(seed)
+    return torch.rand(_IND_ROWS, 16)
+
+
+def _save_indivisible(tmpdir: str) -> None:
+    from torchsnapshot_amd import Snapshot
+
+    st = _make_indivisible(seed=42)
+    Snapshot.take(os.path.join(tmpdir, "snap"), {"obj": _Holder(st)})
+
+
+def _restore_indivisible(tmpdir: str) -> None:
+    from torchsnapshot_amd import Snapshot
+
+    st = _make_indivisible(seed=777)  # wrong values, right layout
+    holder = _Holder(st)
+    Snapshot(os.path.join(tmpdir, "snap")).restore({"obj": holder})
+    full = _indivisible_reference(42)
+    for shard in holder.st.local_shards():
+        lo = shard.metadata.shard_offsets[0]
+        rows = shard.tensor.shape[0]
+        assert torch.equal(shard.tensor, full[lo : lo + rows])
+
+
+def test_sharded_indivisible_save3_restore5():
+    """47 rows over 3 writers (16/16/15) restored over 5 readers
+    (10/10/10/10/7): bit-exact (VERDICT round-1, next-round item 2)."""
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(3, _save_indivisible, d)
+        run_multiprocess(5, _restore_indivisible, d)
+
+
+def test_sharded_indivisible_save5_restore3():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(5, _save_indivisible, d)
+        run_multiprocess(3, _restore_indivisible, d)
