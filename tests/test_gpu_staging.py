@@ -243,16 +243,30 @@ def test_dtensor_on_gpu_single_rank():
             dist.destroy_process_group()
 
 
-def test_stage_flipped_tensor():
-    """Negative-stride views (flip) are materialized before packing."""
-    t = torch.arange(64, dtype=torch.float32, device="cuda").reshape(8, 8)
-    f = torch.flip(t, dims=[0])  # flip returns a copy (contiguous) in torch
-    v = t.as_strided((8, 8), (-8 + 16, 1), storage_offset=0) if False else f
+def test_stage_large_noncontig_materialized():
+    """Non-contiguous tensors >=2 GiB exceed the pack kernel's u32
+    within-tensor byte indexing and must be materialized (contiguous copy)
+    before staging — the engine's only remaining materialization branch
+    (torch itself forbids negative strides, so no flip case exists).
+    Exercises ops/staging.py's size guard on device."""
+    rows = 32768
+    cols = 32772  # rows*cols*2 B = 2 GiB + 256 KiB > 2**31
+    t = torch.empty(cols, rows, dtype=torch.bfloat16, device="cuda")
+    torch.manual_seed(3)
+    t.view(torch.int16).random_()
+    v = t.t()  # non-contiguous view over > 2**31 bytes
+    assert v.numel() * v.element_size() >= 2**31 and not v.is_contiguous()
     engine = staging.get_staging_engine(t.device)
-    batch = engine.stage([f])
+    batch = engine.stage([v])
     batch.wait()
-    assert bytes(batch.memoryview_of(0)) == _ref_bytes(f)
+    got = torch.frombuffer(
+        bytearray(batch.memoryview_of(0)), dtype=torch.bfloat16
+    ).view(rows, cols)
     batch.release()
+    ref = v.contiguous().cpu().view(torch.bfloat16)
+    assert torch.equal(got.view(torch.int16), ref.view(torch.int16))
+    del t, v, ref
+    torch.cuda.empty_cache()
 
 
 def test_gpu_saved_cpu_restored():

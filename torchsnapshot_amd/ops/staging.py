@@ -391,10 +391,10 @@ class StagingEngine:
         The caller must ensure the tensors' producing stream is
         torch.cuda.current_stream() of this thread (true for checkpointing:
         tensors are live parameters/opt states, already materialized)."""
-        # the pack kernel indexes within-tensor bytes as u32 and walks
-        # non-negative strides; materialize the rare exceptions first:
-        # non-contiguous tensors over 2 GiB (chunking splits along dim 0
-        # upstream) and negative-stride views (flips)
+        # the pack kernel indexes within-tensor bytes as u32: materialize
+        # non-contiguous tensors over 2 GiB first (chunking splits along
+        # dim 0 upstream, so this is rare). The stride>=0 check is
+        # defensive only — torch itself forbids negative strides.
         tensors = [
             t
             if (

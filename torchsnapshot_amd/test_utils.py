@@ -169,6 +169,74 @@ def run_multiprocess(nproc: int, fn: Callable[..., None], *args: Any) -> None:
     )
 
 
+def _mp_entry_gpu(
+    rank: int,
+    world_size: int,
+    port: int,
+    backend: str,
+    share_device: bool,
+    module_name: str,
+    qualname: str,
+    args: tuple,
+) -> None:
+    import importlib
+
+    device_index = 0 if share_device else rank
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["LOCAL_RANK"] = str(device_index)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.cuda.set_device(device_index)
+    kwargs: Dict[str, Any] = {}
+    if backend == "nccl":
+        kwargs["device_id"] = torch.device("cuda", device_index)
+    dist.init_process_group(
+        backend=backend, rank=rank, world_size=world_size, **kwargs
+    )
+    try:
+        module = importlib.import_module(module_name)
+        obj: Any = module
+        for part in qualname.split("."):
+            obj = getattr(obj, part)
+        fn = getattr(obj, "__wrapped__", obj)
+        fn(*args)
+    finally:
+        dist.destroy_process_group()
+
+
+def run_multiprocess_gpu(
+    nproc: int,
+    fn: Callable[..., None],
+    *args: Any,
+    backend: str = "nccl",
+    share_device: bool = False,
+) -> None:
+    """GPU flavor of :func:`run_multiprocess`: rank r uses cuda:r (RCCL
+    over xGMI on a multi-GPU node). With ``share_device=True`` every rank
+    uses cuda:0 and the backend must be gloo (RCCL refuses multiple ranks
+    on one device) — this exercises the multi-rank save/restore logic
+    with device tensors even on a 1-GPU box."""
+    if share_device and backend == "nccl":
+        raise ValueError("share_device requires the gloo backend")
+    port = _find_free_port()
+    mp.start_processes(
+        _mp_entry_gpu,
+        args=(
+            nproc,
+            port,
+            backend,
+            share_device,
+            fn.__module__,
+            fn.__qualname__,
+            args,
+        ),
+        nprocs=nproc,
+        start_method="spawn",
+        join=True,
+    )
+
+
 def run_with_dist(nproc: int):
     """Decorator form: the wrapped test launches itself in N gloo-connected
     processes. The test function must be module-level (re-importable)."""
