@@ -71,7 +71,9 @@ class _SigV4:
         )
         canonical_query = "&".join(
             f"{k}={urllib.parse.quote(v, safe='')}"
-            for k, v in sorted(urllib.parse.parse_qsl(parsed.query))
+            for k, v in sorted(
+                urllib.parse.parse_qsl(parsed.query, keep_blank_values=True)
+            )
         )
         canonical_request = "\n".join(
             [
@@ -159,11 +161,21 @@ class S3StoragePlugin(StoragePlugin):
         return sess
 
     async def write(self, write_io: WriteIO) -> None:
-        url = self._url(write_io.path)
-        headers = self.signer.sign("PUT", url, _UNSIGNED)
         mv = memoryview(write_io.buf)
         if mv.format != "B":
             mv = mv.cast("B")
+        if mv.nbytes >= self._multipart_threshold():
+            await self._write_multipart(write_io.path, mv)
+            return
+        if mv.nbytes > 5 * 1024**3:
+            # S3 caps single PUTs at 5 GiB; only reachable if the
+            # multipart threshold was overridden above that
+            raise ValueError(
+                f"S3 single PUT of {mv.nbytes} bytes exceeds the 5 GiB "
+                "limit; lower TSAMD_S3_MULTIPART_THRESHOLD_BYTES"
+            )
+        url = self._url(write_io.path)
+        headers = self.signer.sign("PUT", url, _UNSIGNED)
         headers["content-length"] = str(mv.nbytes)
         sess = await self._session()
         for attempt in range(4):
@@ -181,6 +193,114 @@ class S3StoragePlugin(StoragePlugin):
                     raise
             await asyncio.sleep(0.5 * 2**attempt)
         raise RuntimeError(f"S3 PUT {write_io.path}: retries exhausted")
+
+    @staticmethod
+    def _multipart_threshold() -> int:
+        """Writes at or above this size use multipart upload (the single
+        PUT path caps at S3's 5 GiB limit)."""
+        return int(
+            float(
+                os.environ.get(
+                    "TSAMD_S3_MULTIPART_THRESHOLD_BYTES", str(1024**3)
+                )
+            )
+        )
+
+    @staticmethod
+    def _part_size() -> int:
+        return int(
+            float(os.environ.get("TSAMD_S3_PART_BYTES", str(256 * 1024**2)))
+        )
+
+    async def _mp_request(
+        self, method: str, url: str, payload_hash: str, data=None, ok=(200,)
+    ):
+        """One signed request with the standard retry policy; returns the
+        response (status checked against ``ok``) with body text loaded."""
+        headers = self.signer.sign(method, url, payload_hash)
+        if data is not None:
+            headers["content-length"] = str(memoryview(data).nbytes)
+        sess = await self._session()
+        for attempt in range(4):
+            try:
+                async with sess.request(
+                    method, url, data=data, headers=headers
+                ) as resp:
+                    body = await resp.text()
+                    if resp.status in ok:
+                        return resp.status, body, dict(resp.headers)
+                    if resp.status < 500 and resp.status != 429:
+                        raise RuntimeError(
+                            f"S3 {method} {url.split('?')[0]}: "
+                            f"{resp.status} {body[:500]}"
+                        )
+            except (OSError, asyncio.TimeoutError):
+                if attempt == 3:
+                    raise
+            await asyncio.sleep(0.5 * 2**attempt)
+        raise RuntimeError(f"S3 {method}: retries exhausted")
+
+    async def _write_multipart(self, path: str, mv: memoryview) -> None:
+        """Multipart upload: initiate -> concurrent part PUTs -> complete;
+        aborted on failure so no orphaned parts accrue charges."""
+        import re
+
+        base = self._url(path)
+        _, body, _ = await self._mp_request(
+            "POST", f"{base}?uploads=", _EMPTY_SHA256
+        )
+        m = re.search(r"<UploadId>([^<]+)</UploadId>", body)
+        if not m:
+            raise RuntimeError(f"S3 initiate multipart: no UploadId in {body[:300]}")
+        upload_id = m.group(1)
+        part_size = self._part_size()
+        # S3 allows at most 10k parts; grow the part size if needed
+        min_part = -(-mv.nbytes // 10000)
+        part_size = max(part_size, min_part, 5 * 1024 * 1024)
+        n_parts = -(-mv.nbytes // part_size)
+        etags: Dict[int, str] = {}
+        sem = asyncio.Semaphore(4)
+
+        async def put_part(idx: int) -> None:
+            lo = idx * part_size
+            hi = min(lo + part_size, mv.nbytes)
+            url = (
+                f"{base}?partNumber={idx + 1}"
+                f"&uploadId={urllib.parse.quote(upload_id, safe='')}"
+            )
+            async with sem:
+                _, _, hdrs = await self._mp_request(
+                    "PUT", url, _UNSIGNED, data=mv[lo:hi]
+                )
+            etags[idx + 1] = hdrs.get("ETag", hdrs.get("Etag", ""))
+
+        try:
+            await asyncio.gather(*(put_part(i) for i in range(n_parts)))
+            xml = (
+                "<CompleteMultipartUpload>"
+                + "".join(
+                    f"<Part><PartNumber>{n}</PartNumber>"
+                    f"<ETag>{etags[n]}</ETag></Part>"
+                    for n in sorted(etags)
+                )
+                + "</CompleteMultipartUpload>"
+            ).encode()
+            url = f"{base}?uploadId={urllib.parse.quote(upload_id, safe='')}"
+            status, body, _ = await self._mp_request(
+                "POST", url, hashlib.sha256(xml).hexdigest(), data=xml
+            )
+            # S3 can return 200 with an <Error> body for completes
+            if "<Error>" in body:
+                raise RuntimeError(f"S3 complete multipart: {body[:500]}")
+        except BaseException:
+            try:
+                url = f"{base}?uploadId={urllib.parse.quote(upload_id, safe='')}"
+                await self._mp_request(
+                    "DELETE", url, _EMPTY_SHA256, ok=(200, 204)
+                )
+            except Exception:
+                pass  # abort is best-effort; the original error wins
+            raise
 
     async def read(self, read_io: ReadIO) -> None:
         url = self._url(read_io.path)
