@@ -159,3 +159,88 @@ def test_sharded_indivisible_save5_restore3():
     with tempfile.TemporaryDirectory() as d:
         run_multiprocess(5, _save_indivisible, d)
         run_multiprocess(3, _restore_indivisible, d)
+
+
+# ---------------------------------------------------------------------------
+# embedding-table workload: multiple row-wise sharded tables, save on 4
+# ranks, reshard-restore on 2 (BASELINE.json config 5 shape; mirror of
+# reference tests/gpu_tests/test_torchrec.py:200-306 reshard matrix)
+# ---------------------------------------------------------------------------
+
+_N_TABLES = 3
+
+
+def _emb_row_values(table_idx: int, rows: torch.Tensor) -> torch.Tensor:
+    return (
+        ((rows.to(torch.float64) * 2654435761.0 + table_idx * 97.0) % 1000003.0)
+        / 1000003.0
+    ).to(torch.float32)
+
+
+def _make_tables(zero: bool = False):
+    from torch.distributed._shard import sharded_tensor
+    from torch.distributed._shard.sharding_spec import ChunkShardingSpec
+
+    world_size = dist.get_world_size()
+    spec = ChunkShardingSpec(
+        dim=0, placements=[f"rank:{r}/cpu" for r in range(world_size)]
+    )
+    tables = {}
+    for i in range(_N_TABLES):
+        st = sharded_tensor.empty(spec, (50 + i * 7, 8))  # uneven everywhere
+        for shard in st.local_shards():
+            off = shard.metadata.shard_offsets[0]
+            n = shard.tensor.shape[0]
+            if zero:
+                shard.tensor.zero_()
+            else:
+                shard.tensor.copy_(
+                    _emb_row_values(i, torch.arange(off, off + n))
+                    .unsqueeze(1)
+                    .expand(n, 8)
+                )
+        tables[f"table_{i}"] = st
+    return tables
+
+
+class _Tables:
+    def __init__(self, tables):
+        self.tables = tables
+
+    def state_dict(self):
+        return dict(self.tables)
+
+    def load_state_dict(self, sd):
+        self.tables = dict(sd)
+
+
+def _emb_save(tmpdir: str) -> None:
+    from torchsnapshot_amd import Snapshot
+
+    Snapshot.take(
+        os.path.join(tmpdir, "snap"), {"emb": _Tables(_make_tables())}
+    )
+
+
+def _emb_restore(tmpdir: str) -> None:
+    from torchsnapshot_amd import Snapshot
+
+    holder = _Tables(_make_tables(zero=True))
+    Snapshot(os.path.join(tmpdir, "snap")).restore({"emb": holder})
+    for i in range(_N_TABLES):
+        st = holder.tables[f"table_{i}"]
+        for shard in st.local_shards():
+            off = shard.metadata.shard_offsets[0]
+            n = shard.tensor.shape[0]
+            want = (
+                _emb_row_values(i, torch.arange(off, off + n))
+                .unsqueeze(1)
+                .expand(n, 8)
+            )
+            assert torch.equal(shard.tensor, want), (i, off, n)
+
+
+def test_embedding_tables_save4_reshard_restore2():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(4, _emb_save, d)
+        run_multiprocess(2, _emb_restore, d)
