@@ -380,3 +380,39 @@ def test_tied_weights_batched():
         snap.restore({"m": m2})
         assert torch.equal(m2.a.weight, m.a.weight)
         assert torch.equal(m2.c.weight, m.c.weight)
+
+
+def test_parallel_segment_fs_io(monkeypatch):
+    """Large files go through the concurrent-segment fs path; round-trip
+    stays bit-exact (cold-read tuning, VERDICT round-1 item 4)."""
+    import torch
+
+    from torchsnapshot_amd import Snapshot, StateDict
+    from torchsnapshot_amd.test_utils import tmp_snapshot_path
+
+    monkeypatch.setenv("TSAMD_FS_PARALLEL_IO_MIN_BYTES", str(64 * 1024))
+    monkeypatch.setenv("TSAMD_FS_IO_SEGMENT_BYTES", str(17 * 1024))  # odd size
+    sd = StateDict(big=torch.rand(512, 257), small=torch.rand(3))
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict()
+        snap.restore({"sd": out})
+        assert torch.equal(out["big"], sd["big"])
+        assert torch.equal(out["small"], sd["small"])
+
+
+def test_parallel_segment_fs_io_fsync(monkeypatch):
+    import torch
+
+    from torchsnapshot_amd import Snapshot, StateDict
+    from torchsnapshot_amd.test_utils import tmp_snapshot_path
+
+    monkeypatch.setenv("TSAMD_FS_PARALLEL_IO_MIN_BYTES", str(64 * 1024))
+    monkeypatch.setenv("TSAMD_FS_IO_SEGMENT_BYTES", str(32 * 1024))
+    monkeypatch.setenv("TSAMD_FSYNC", "1")
+    sd = StateDict(big=torch.rand(300, 300))
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict()
+        snap.restore({"sd": out})
+        assert torch.equal(out["big"], sd["big"])

@@ -119,6 +119,18 @@ def get_storage_write_chunk_bytes() -> int:
     return _env_bytes("TSAMD_FS_WRITE_CHUNK_BYTES", 256 * _MB)
 
 
+def get_fs_parallel_io_min_bytes() -> int:
+    """Files at/above this size are read/written as concurrent segments
+    (multiple NVMe queues per file; closes the cold-read gap vs raw
+    device bandwidth that a single-stream pread leaves)."""
+    return _env_bytes("TSAMD_FS_PARALLEL_IO_MIN_BYTES", 128 * _MB)
+
+
+def get_fs_io_segment_bytes() -> int:
+    """Segment size for parallel per-file I/O."""
+    return _env_bytes("TSAMD_FS_IO_SEGMENT_BYTES", 64 * _MB)
+
+
 # -- elasticity --------------------------------------------------------------
 
 def is_sharded_elasticity_root_only() -> bool:
