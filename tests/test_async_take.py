@@ -174,3 +174,21 @@ def test_async_take_peer_failure_world2():
 
     with tf.TemporaryDirectory() as d:
         run_multiprocess(2, _async_take_peer_failure, d)
+
+
+def test_async_take_custom_tensor_prepare_func():
+    """async_take accepts the save-time tensor transform too (parity with
+    reference snapshot.py:230-293)."""
+    sd = StateDict(w=torch.rand(64, 32))
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "snap")
+        pending = Snapshot.async_take(
+            path,
+            {"sd": sd},
+            _custom_tensor_prepare_func=lambda p, t: t.to(torch.bfloat16),
+        )
+        snap = pending.wait()
+        assert snap.get_manifest()["0/sd/w"]["dtype"] == "bfloat16"
+        out = StateDict(w=torch.zeros(64, 32))
+        snap.restore({"sd": out})
+        assert torch.equal(out["w"], sd["w"].to(torch.bfloat16).float())

@@ -227,6 +227,7 @@ class Snapshot:
         pg: Optional[dist.ProcessGroup] = None,
         replicated: Optional[List[str]] = None,
         storage_options: Optional[Dict[str, Any]] = None,
+        _custom_tensor_prepare_func: Optional[Any] = None,
     ) -> "PendingSnapshot":
         torch._C._log_api_usage_once("torchsnapshot_amd.Snapshot.async_take")
         cls._validate_app_state(app_state)
@@ -261,6 +262,7 @@ class Snapshot:
                 pg_wrapper=pg_wrapper,
                 replicated=replicated,
                 is_async=True,
+                custom_tensor_prepare_func=_custom_tensor_prepare_func,
             )
             # once staging is complete, the app may mutate its state freely
             pending_io_work.wait_staged()
