@@ -104,6 +104,10 @@ class FSStoragePlugin(StoragePlugin):
                 os.posix_fadvise(
                     fd, file_off, len(mv), os.POSIX_FADV_SEQUENTIAL
                 )
+                if os.environ.get("TSAMD_FS_WILLNEED", "0") not in ("0", ""):
+                    os.posix_fadvise(
+                        fd, file_off, len(mv), os.POSIX_FADV_WILLNEED
+                    )
             except (AttributeError, OSError):
                 pass
             nbytes = len(mv)
