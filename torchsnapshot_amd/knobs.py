@@ -120,14 +120,15 @@ def get_storage_write_chunk_bytes() -> int:
 
 
 def get_fs_parallel_io_min_bytes() -> int:
-    """Files at/above this size are read/written as concurrent segments
-    (multiple NVMe queues per file; closes the cold-read gap vs raw
-    device bandwidth that a single-stream pread leaves)."""
+    """Files at/above this size are READ as concurrent segments (multiple
+    NVMe queues per file; closes the cold-read gap vs raw device bandwidth
+    that a single-stream pread leaves). Writes stay single-stream per file
+    — buffered writes serialize on the inode lock anyway."""
     return _env_bytes("TSAMD_FS_PARALLEL_IO_MIN_BYTES", 128 * _MB)
 
 
 def get_fs_io_segment_bytes() -> int:
-    """Segment size for parallel per-file I/O."""
+    """Segment size for parallel per-file reads."""
     return _env_bytes("TSAMD_FS_IO_SEGMENT_BYTES", 64 * _MB)
 
 

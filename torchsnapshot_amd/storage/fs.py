@@ -2,14 +2,16 @@
 
 Design (MI355X node, local NVMe target): blocking pwrite/pread on raw fds in
 a worker-thread pool — the syscalls release the GIL, so N threads drive N
-NVMe queues concurrently. Writes are large and sequential (the batcher packs
-small tensors into multi-hundred-MB slabs upstream), which is the layout
-NVMe likes. Files above ``TSAMD_FS_PARALLEL_IO_MIN_BYTES`` are additionally
-read/written as concurrent disjoint segments, each with
-POSIX_FADV_SEQUENTIAL readahead hints — a single-stream pread is readahead-
-window-bound and leaves ~20% of raw NVMe bandwidth on a cold read
-(profiles/r01_measurements.md). No asyncio file libraries are used (parity
-of behavior with the reference's aiofiles plugin,
+NVMe queues concurrently. Writes are large, sequential, ONE stream per file
+(buffered writes to one inode serialize on i_rwsem, so intra-file write
+parallelism only costs cross-file concurrency — measured, see
+profiles/r02_measurements.md); the batcher packs small tensors into
+multi-hundred-MB slabs upstream, which is the layout NVMe likes. Reads of
+files above ``TSAMD_FS_PARALLEL_IO_MIN_BYTES`` fan out into concurrent
+disjoint segments with SEQUENTIAL+WILLNEED fadvise readahead — a
+single-stream pread is readahead-window-bound and leaves ~20% of raw NVMe
+bandwidth on a cold read. No asyncio file libraries are used (parity of
+behavior with the reference's aiofiles plugin,
 torchsnapshot/storage_plugins/fs.py:28-51, via a different mechanism).
 """
 
@@ -87,12 +89,6 @@ class FSStoragePlugin(StoragePlugin):
         finally:
             os.close(fd)
 
-    def _pwrite_segment(self, fd: int, mv: memoryview, off: int) -> None:
-        total = len(mv)
-        done = 0
-        while done < total:
-            done += os.pwrite(fd, mv[done:], off + done)
-
     def _pread_segment(
         self, full: str, mv: memoryview, file_off: int
     ) -> None:
@@ -104,7 +100,7 @@ class FSStoragePlugin(StoragePlugin):
                 os.posix_fadvise(
                     fd, file_off, len(mv), os.POSIX_FADV_SEQUENTIAL
                 )
-                if os.environ.get("TSAMD_FS_WILLNEED", "0") not in ("0", ""):
+                if os.environ.get("TSAMD_FS_WILLNEED", "1") not in ("0", ""):
                     os.posix_fadvise(
                         fd, file_off, len(mv), os.POSIX_FADV_WILLNEED
                     )
@@ -145,32 +141,17 @@ class FSStoragePlugin(StoragePlugin):
     # -- StoragePlugin ------------------------------------------------------
 
     async def write(self, write_io: WriteIO) -> None:
+        # one sequential stream per file: ext4/xfs serialize buffered
+        # writes to one inode on i_rwsem, so intra-file write parallelism
+        # only steals executor threads from OTHER files (measured: 10-step
+        # sustained save fell 9.9 -> 5.8 GB/s with segmented writes, and
+        # durable saves didn't improve — profiles/r02_measurements.md).
+        # Cross-file concurrency comes from the scheduler's many in-flight
+        # write requests.
         loop = asyncio.get_running_loop()
-        mv = _as_u8_mv(write_io.buf)
-        if mv.nbytes < knobs.get_fs_parallel_io_min_bytes():
-            await loop.run_in_executor(
-                self._executor, self._write_sync, write_io.path, write_io.buf
-            )
-            return
-        # large file: concurrent disjoint pwrite segments on one fd
-        full = self._abspath(write_io.path)
-        self._ensure_dir(os.path.dirname(full))
-        fd = os.open(full, os.O_WRONLY | os.O_CREAT | os.O_TRUNC, 0o644)
-        try:
-            await asyncio.gather(
-                *(
-                    loop.run_in_executor(
-                        self._executor, self._pwrite_segment, fd, mv[s:e], s
-                    )
-                    for s, e in self._segments(mv.nbytes)
-                )
-            )
-            if _fsync_enabled():
-                await loop.run_in_executor(
-                    self._executor, self._fsync_file_and_dir, fd, full
-                )
-        finally:
-            os.close(fd)
+        await loop.run_in_executor(
+            self._executor, self._write_sync, write_io.path, write_io.buf
+        )
 
     async def read(self, read_io: ReadIO) -> None:
         loop = asyncio.get_running_loop()
