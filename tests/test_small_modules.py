@@ -148,3 +148,57 @@ def test_cli_inspect_and_cat(tmp_path, capsys):
     assert "0/sd/w" in out and "tensor" in out
     assert main(["cat", p, "0/sd/n"]) == 0
     assert capsys.readouterr().out.strip() == "9"
+
+
+# ---------------------------------------------------------------------------
+# equality oracles (mirror of reference tests/test_test_utils.py)
+# ---------------------------------------------------------------------------
+
+
+def test_check_state_dict_eq_plain():
+    import torch
+
+    from torchsnapshot_amd.test_utils import (
+        assert_state_dict_eq,
+        check_state_dict_eq,
+    )
+
+    a = {"x": torch.ones(4), "n": 3, "nested": {"y": torch.zeros(2)}}
+    b = {"x": torch.ones(4), "n": 3, "nested": {"y": torch.zeros(2)}}
+    assert check_state_dict_eq(a, b)
+    b["x"] = torch.zeros(4)
+    assert not check_state_dict_eq(a, b)
+    import unittest
+
+    with pytest.raises(AssertionError):
+        assert_state_dict_eq(unittest.TestCase(), a, b)
+
+
+def test_check_state_dict_eq_dtype_and_shape_mismatch():
+    import torch
+
+    from torchsnapshot_amd.test_utils import check_state_dict_eq
+
+    assert not check_state_dict_eq(
+        {"x": torch.ones(4)}, {"x": torch.ones(4, dtype=torch.float64)}
+    )
+    assert not check_state_dict_eq({"x": torch.ones(4)}, {"x": torch.ones(5)})
+    assert not check_state_dict_eq({"x": torch.ones(4)}, {})
+
+
+def test_rand_tensor_dtypes():
+    import torch
+
+    from torchsnapshot_amd.test_utils import rand_tensor
+
+    for dtype in (
+        torch.float32,
+        torch.bfloat16,
+        torch.int64,
+        torch.bool,
+        torch.complex64,
+    ):
+        t = rand_tensor((4, 4), dtype)
+        assert t.dtype == dtype and t.shape == (4, 4)
+    q = rand_tensor((8,), torch.qint8)
+    assert q.is_quantized
