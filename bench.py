@@ -155,10 +155,20 @@ def main() -> None:
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     use_cuda = args.device == "cuda"
+    # rehearsal escape hatch for 1-GPU boxes: TSAMD_BENCH_SHARE_DEVICE=1
+    # puts every rank on cuda:0 over gloo so the multi-rank DTensor save
+    # path can be exercised before an 8-GPU node exists. The driver's
+    # launch (no env set) is unaffected: N ranks, cuda:LOCAL_RANK, RCCL.
+    share_device = os.environ.get("TSAMD_BENCH_SHARE_DEVICE", "0") not in (
+        "0",
+        "",
+    )
+    if share_device:
+        local_rank = 0
     if world_size > 1:
         import torch.distributed as dist
 
-        if use_cuda:
+        if use_cuda and not share_device:
             torch.cuda.set_device(local_rank)
             dist.init_process_group(
                 backend="nccl", device_id=torch.device("cuda", local_rank)
