@@ -55,3 +55,27 @@ def test_example_ddp_world2():
     )
     assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
     assert "restored; step = 1" in r.stdout
+
+
+def test_example_sharded_embedding_world2():
+    r = subprocess.run(
+        [
+            sys.executable,
+            "-m",
+            "torch.distributed.run",
+            "--nnodes=1",
+            "--nproc-per-node",
+            "2",
+            "--master-addr",
+            "127.0.0.1",
+            "--master-port",
+            "29572",
+            os.path.join(REPO, "examples", "sharded_embedding.py"),
+        ],
+        cwd=REPO,
+        capture_output=True,
+        text=True,
+        timeout=240,
+    )
+    assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
+    assert "restored; all table rows verified" in r.stdout
