@@ -274,3 +274,42 @@ def test_s3_oversize_single_put_guarded(monkeypatch):
 
     with pytest.raises(ValueError, match="5 GiB"):
         _asyncio.run(go())
+
+
+def test_s3_multipart_abort_on_failure(fake_s3, monkeypatch):
+    """A failed part upload aborts the multipart upload (no orphaned
+    parts) and surfaces the error."""
+    import asyncio as _asyncio
+
+    from torchsnapshot_amd.io_types import WriteIO
+    from torchsnapshot_amd.storage.s3 import S3StoragePlugin
+
+    # parts have a hard 5 MiB floor (S3 minimum), so use a 12 MiB object
+    monkeypatch.setenv("TSAMD_S3_MULTIPART_THRESHOLD_BYTES", str(1024 * 1024))
+    monkeypatch.setenv("TSAMD_S3_PART_BYTES", str(5 * 1024 * 1024))
+    plugin = S3StoragePlugin("bkt/abort", _options(fake_s3))
+
+    orig = plugin._mp_request
+    calls = {"n": 0}
+
+    async def flaky(method, url, payload_hash, data=None, ok=(200,)):
+        if method == "PUT" and "partNumber=2" in url:
+            calls["n"] += 1
+            raise RuntimeError("injected part failure")
+        return await orig(method, url, payload_hash, data=data, ok=ok)
+
+    plugin._mp_request = flaky
+
+    async def go():
+        try:
+            await plugin.write(
+                WriteIO(path="obj", buf=bytearray(12 * 1024 * 1024))
+            )
+        finally:
+            await plugin.close()
+
+    with pytest.raises(RuntimeError, match="injected part failure"):
+        _asyncio.run(go())
+    assert calls["n"] >= 1
+    assert not fake_s3.uploads, "multipart upload was not aborted"
+    assert "bkt/abort/obj" not in fake_s3.objects
