@@ -314,3 +314,102 @@ def test_dtensor_world8_reshard_to_world4():
     with tempfile.TemporaryDirectory() as d:
         run_multiprocess_gpu(8, _dtensor_save_n, d)
         run_multiprocess_gpu(4, _dtensor_restore_n, d)
+
+
+# ---------------------------------------------------------------------------
+# advisor-regression scenarios with DEVICE tensors (shared GPU, world 2):
+# the round-1 restore-breaking bugs involved the partitioner/batcher
+# interplay, which has a different code path (HIP slab engine) on device
+# ---------------------------------------------------------------------------
+
+
+class _TiedState:
+    def __init__(self, shared, filler):
+        self.shared = shared
+        self.filler = filler
+
+    def state_dict(self):
+        return {"a": self.shared, "b": self.shared, "filler": self.filler}
+
+    def load_state_dict(self, sd):
+        self.shared = sd["a"]
+        self.filler = sd["filler"]
+
+
+def _tied_device_save_restore(tmpdir: str) -> None:
+    rank = dist.get_rank()
+    torch.manual_seed(7)
+    shared = torch.rand(1024, device="cuda")
+    filler = (
+        torch.rand(65536, device="cuda")
+        if rank == 0
+        else torch.rand(16, device="cuda")
+    )
+    path = os.path.join(tmpdir, "snap")
+    state = _TiedState(shared, filler)
+    Snapshot.take(path, {"m": state}, replicated=["m/a", "m/b"])
+
+    target = _TiedState(
+        torch.zeros(1024, device="cuda"), torch.zeros_like(filler)
+    )
+    Snapshot(path).restore({"m": target})
+    assert torch.equal(target.shared, shared)
+
+
+def test_tied_replicated_device_world2_shared_gpu():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess_gpu(
+            2, _tied_device_save_restore, d, backend="gloo", share_device=True
+        )
+
+
+def _repl_dtensor_device_pieces(tmpdir: str) -> None:
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import DTensor
+    from torch.distributed.tensor.placement_types import Replicate
+
+    os.environ["TSAMD_MAX_SHARD_SIZE_BYTES"] = "2048"
+    try:
+        path = os.path.join(tmpdir, "snap")
+        mesh = init_device_mesh("cpu", (dist.get_world_size(),))
+        torch.manual_seed(3)
+        full = torch.rand(64, 32)
+        # replicated DTensor whose LOCAL tensor lives on the GPU; the
+        # mesh stays cpu/gloo (shared single device)
+        dt = DTensor.from_local(
+            full.cuda(), mesh, [Replicate()], run_check=False
+        )
+
+        class H:
+            def __init__(self, o):
+                self.obj = o
+
+            def state_dict(self):
+                return {"t": self.obj}
+
+            def load_state_dict(self, sd):
+                self.obj = sd["t"]
+
+        Snapshot.take(path, {"m": H(dt)})
+        dt2 = DTensor.from_local(
+            torch.zeros(64, 32, device="cuda"),
+            mesh,
+            [Replicate()],
+            run_check=False,
+        )
+        holder = H(dt2)
+        Snapshot(path).restore({"m": holder})
+        assert torch.equal(holder.obj.to_local().cpu(), full)
+    finally:
+        del os.environ["TSAMD_MAX_SHARD_SIZE_BYTES"]
+
+
+def test_replicated_dtensor_device_pieces_world2_shared_gpu():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess_gpu(
+            2,
+            _repl_dtensor_device_pieces,
+            d,
+            backend="gloo",
+            share_device=True,
+        )
