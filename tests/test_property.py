@@ -153,3 +153,50 @@ def test_manifest_entry_json_round_trip(data):
 
     rebuilt = entry_from_dict(json.loads(json.dumps(entry.to_dict())))
     assert rebuilt.to_dict() == entry.to_dict()
+
+
+@given(
+    nbytes=st.integers(min_value=0, max_value=4096),
+    seed=st.integers(min_value=0, max_value=2**31),
+)
+@settings(max_examples=60, deadline=None)
+def test_psum64_subrange_additivity(nbytes, seed):
+    """psum64 with file-global word indexing is additive over disjoint
+    8-aligned subranges — the invariant the batched-slab ranged checksum
+    verification relies on (integrity.verify_ranged_buffer)."""
+    import random
+
+    from torchsnapshot_amd.integrity import psum64_hexdigest
+
+    rng = random.Random(seed)
+    buf = bytes(rng.getrandbits(8) for _ in range(nbytes))
+    whole = psum64_hexdigest(buf)
+
+    # random 8-aligned cut points
+    n_cuts = rng.randint(0, 5)
+    cuts = sorted({rng.randrange(0, nbytes + 1) & ~7 for _ in range(n_cuts)})
+    bounds = [0] + cuts + [nbytes]
+    total = 0
+    for lo, hi in zip(bounds, bounds[1:]):
+        if hi <= lo:
+            continue
+        part = psum64_hexdigest(buf[lo:hi], word_base=lo // 8)
+        total = (total + int(part[len("psum64:"):], 16)) % (1 << 64)
+    assert "psum64:" + format(total, "016x") == whole
+
+
+@given(
+    pad=st.integers(min_value=0, max_value=64),
+    seed=st.integers(min_value=0, max_value=2**31),
+)
+@settings(max_examples=30, deadline=None)
+def test_psum64_zero_padding_invariant(pad, seed):
+    """Zero bytes contribute nothing: the checksum of a buffer equals the
+    checksum of the buffer + trailing zeros (slab padding invariant)."""
+    import random
+
+    from torchsnapshot_amd.integrity import psum64_hexdigest
+
+    rng = random.Random(seed)
+    buf = bytes(rng.getrandbits(8) for _ in range(256))
+    assert psum64_hexdigest(buf) == psum64_hexdigest(buf + b"\0" * pad)
