@@ -86,18 +86,11 @@ class FSStoragePlugin(StoragePlugin):
                 off += os.pwrite(fd, mv[off : off + chunk], off)
             if _fsync_enabled():
                 self._fsync_file_and_dir(fd, full)
-            elif os.environ.get("TSAMD_FS_WRITEBACK_HINT", "0") not in (
-                "0",
-                "",
-            ):
-                # experiment knob: kick off async writeback + drop the
-                # cached pages now, so repeated saves never accumulate a
-                # dirty backlog that throttles later steps. Costs warm
-                # restore (reads become cold). Off by default.
-                try:
-                    os.posix_fadvise(fd, 0, 0, os.POSIX_FADV_DONTNEED)
-                except (AttributeError, OSError):
-                    pass
+            # note: fadvise(DONTNEED) after payload writes was measured
+            # and rejected — it did not lift the sustained save rate
+            # (9.50 vs 9.40 GB/s, writeback equilibrium dominates either
+            # way) and it cost warm restores 3x (50 -> 16 GB/s).
+            # profiles/r02_measurements.md.
         finally:
             os.close(fd)
 
