@@ -127,17 +127,30 @@ def member_key(path: str, start: int, end: int) -> str:
     return f"{path}#{start}-{end}"
 
 
+def len_key(path: str) -> str:
+    """Checksum-file key recording a payload file's byte length (lets
+    restore verify a file read as byte-range tiles once the tiles cover
+    the whole file)."""
+    return f"{path}#len"
+
+
+def psum64_value(buf, word_base: int = 0) -> int:
+    """psum64 as an integer (for accumulating partial sums)."""
+    return int(psum64_hexdigest(buf, word_base)[len("psum64:"):], 16)
+
+
 def verify_ranged_buffer(
     path: str, buf, byte_range, expected: Dict[str, str]
-) -> None:
+) -> bool:
     """Verify a byte-range read (a batched-slab member or a merged span of
     members) against per-member psum64 values recorded at save time.
+    Returns True when verification happened, False when the range had no
+    recorded members (the caller may then fall back to whole-file
+    partial-sum accumulation); raises on mismatch.
 
     psum64 is additive over disjoint file word-ranges and slab padding is
     zeroed, so the expected checksum of any member-aligned span is the
-    (mod 2^64) sum of the recorded per-member values inside it. Ranges not
-    covered by any recorded member (e.g. tiled reads of an unbatched
-    tensor, or snapshots saved without checksumming) are skipped.
+    (mod 2^64) sum of the recorded per-member values inside it.
     """
     start, end = byte_range
     prefix = path + "#"
@@ -147,12 +160,14 @@ def verify_ranged_buffer(
         if not k.startswith(prefix) or not v.startswith("psum64:"):
             continue
         s_str, _, e_str = k[len(prefix):].partition("-")
+        if not (s_str.isdigit() and e_str.isdigit()):
+            continue
         s, e = int(s_str), int(e_str)
         if s >= start and e <= end:
             total = (total + int(v[len("psum64:"):], 16)) % (1 << 64)
             found = True
     if not found:
-        return
+        return False
     if start % 8 != 0:
         # slab members are 64-byte aligned; a misaligned span can't use
         # file-global word indexing — don't verify rather than misreport
@@ -160,7 +175,7 @@ def verify_ranged_buffer(
             "skipping checksum verification of misaligned span %s[%d:%d]",
             path, start, end,
         )
-        return
+        return False
     got = psum64_hexdigest(buf, word_base=start // 8)
     want = "psum64:" + format(total, "016x")
     if got != want:
@@ -169,6 +184,7 @@ def verify_ranged_buffer(
             f"recorded {want}, read back {got} — the file is corrupted or "
             "was modified after the snapshot was committed"
         )
+    return True
 
 
 def verify_buffer(path: str, buf, expected: Dict[str, str]) -> None:

@@ -265,6 +265,9 @@ def execute_write_reqs(
                         getattr(req.stager, "member_checksums", None) or []
                     ):
                         checksums[integrity.member_key(req.path, s, e)] = v
+                    # file length, so TILED byte-range reads of this file
+                    # can be verified once their union covers the file
+                    checksums[integrity.len_key(req.path)] = str(nbytes)
                 stats.staged_reqs += 1
                 stats.staged_bytes += nbytes
                 staged_remaining -= 1
@@ -374,6 +377,8 @@ def execute_read_reqs(
             thread_name_prefix="tsamd-consume",
         )
         ctx = StageContext(executor=executor)
+        # path -> [accumulated psum64, [(start, end), ...]] for tiled reads
+        partial_sums: Dict[str, list] = {}
 
         async def handle(req: ReadReq) -> None:
             cost = req.consumer.get_consuming_cost_bytes()
@@ -391,16 +396,35 @@ def execute_read_reqs(
                 buf = read_io.buf
                 stats.io_bytes += memoryview(buf).nbytes
                 if checksums:
+                    loop = asyncio.get_running_loop()
                     if req.byte_range is None:
-                        await asyncio.get_running_loop().run_in_executor(
+                        await loop.run_in_executor(
                             executor, integrity.verify_buffer, req.path, buf,
                             checksums,
                         )
                     else:
-                        await asyncio.get_running_loop().run_in_executor(
+                        verified = await loop.run_in_executor(
                             executor, integrity.verify_ranged_buffer,
                             req.path, buf, req.byte_range, checksums,
                         )
+                        s, e = req.byte_range
+                        whole = checksums.get(req.path, "")
+                        if (
+                            not verified
+                            and whole.startswith("psum64:")
+                            and s % 8 == 0
+                        ):
+                            # tiled read of an unbatched file: accumulate
+                            # the partial sum; once the tiles cover the
+                            # whole file, the total is checked below
+                            val = await loop.run_in_executor(
+                                executor, integrity.psum64_value, buf, s // 8
+                            )
+                            rec = partial_sums.setdefault(
+                                req.path, [0, []]
+                            )
+                            rec[0] = (rec[0] + val) % (1 << 64)
+                            rec[1].append((s, e))
                 t0 = time.monotonic()
                 await req.consumer.consume_buffer(ctx, buf)
                 stats.consume_s += time.monotonic() - t0
@@ -416,6 +440,31 @@ def execute_read_reqs(
             errors = [r for r in results if isinstance(r, BaseException)]
             if errors:
                 raise errors[0]
+            # tiled-read verification: files whose byte-range tiles ended
+            # up covering the whole file get their accumulated psum64
+            # checked against the recorded whole-file value
+            for path, (total, ranges) in partial_sums.items():
+                want = checksums.get(path) if checksums else None
+                want_len = checksums.get(integrity.len_key(path)) if checksums else None
+                if not want or not want_len:
+                    continue
+                ranges.sort()
+                covered = 0
+                for s, e in ranges:
+                    if s != covered:
+                        covered = -1
+                        break
+                    covered = e
+                if covered != int(want_len):
+                    continue  # partial coverage: cannot verify
+                got = "psum64:" + format(total, "016x")
+                if got != want:
+                    raise RuntimeError(
+                        f"checksum mismatch for tiled read of '{path}': "
+                        f"snapshot recorded {want}, read back {got} — the "
+                        "file is corrupted or was modified after the "
+                        "snapshot was committed"
+                    )
         finally:
             reporter.cancel()
             executor.shutdown(wait=False)

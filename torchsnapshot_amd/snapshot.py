@@ -628,8 +628,19 @@ class Snapshot:
                 futs[p] = fut
             read_reqs = _batch_reads(read_reqs)
             budget = get_process_memory_budget_bytes(pg_wrapper)
+            from . import integrity
+
+            checksums = (
+                integrity.load_checksums(storage, self.metadata.world_size)
+                if integrity.verification_enabled()
+                else None
+            )
             sync_execute_read_reqs(
-                read_reqs, storage, budget, rank=pg_wrapper.get_rank()
+                read_reqs,
+                storage,
+                budget,
+                rank=pg_wrapper.get_rank(),
+                checksums=checksums,
             )
             values = {p: f.obj for p, f in futs.items()}
             return inflate(sub_manifest, values, prefix=key)
@@ -685,7 +696,16 @@ class Snapshot:
             )
             read_reqs = _batch_reads(read_reqs)
             budget = memory_budget_bytes or (32 * 1024**3)
-            sync_execute_read_reqs(read_reqs, storage, budget, rank=0)
+            from . import integrity
+
+            checksums = (
+                integrity.load_checksums(storage, self.metadata.world_size)
+                if integrity.verification_enabled()
+                else None
+            )
+            sync_execute_read_reqs(
+                read_reqs, storage, budget, rank=0, checksums=checksums
+            )
             return fut.obj
         finally:
             storage.sync_close()
