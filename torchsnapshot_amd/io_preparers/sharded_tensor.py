@@ -131,6 +131,9 @@ class ShardConsumer(BufferConsumer):
         self.targets = targets
         self._pinned_block = None
         self._pinned_nbytes = 0
+        # (expected psum64 value, word_base) set by the read scheduler
+        # when this consumer verifies on device (will_verify_on_device)
+        self.expected_psum = None
 
     def all_targets_on_device(self) -> bool:
         return bool(self.targets) and all(
@@ -162,6 +165,16 @@ class ShardConsumer(BufferConsumer):
             return self.targets[0][0].device
         return None
 
+    def will_verify_on_device(self) -> bool:
+        from ..ops.staging import HIP_EXT_AVAILABLE
+
+        return (
+            HIP_EXT_AVAILABLE
+            and self._pinned_block is not None
+            and self.all_targets_on_device()
+            and self.shard_entry.serializer != "torch_save"
+        )
+
     def consume_from_device_u8(self, dev_u8: torch.Tensor) -> None:
         dtype = str_to_dtype(self.shard_entry.dtype)
         shard = (
@@ -186,6 +199,12 @@ class ShardConsumer(BufferConsumer):
                 dev_u8 = self._pinned_block.tensor[:n].to(
                     self.targets[0][0].device, non_blocking=False
                 )
+                if self.expected_psum is not None:
+                    from ..ops.staging import verify_device_psum
+
+                    verify_device_psum(
+                        dev_u8, self.expected_psum, self.shard_entry.location
+                    )
                 shard = (
                     dev_u8.view(dtype).reshape(tuple(self.shard_entry.shape))
                     if dtype != torch.uint8

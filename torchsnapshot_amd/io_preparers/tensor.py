@@ -356,6 +356,9 @@ class TensorBufferConsumer(BufferConsumer):
         self.fut = fut
         self._pinned_block = None
         self._pinned_nbytes = 0
+        # (expected psum64 value, word_base) set by the read scheduler
+        # when this consumer verifies on device (will_verify_on_device)
+        self.expected_psum = None
 
     def alloc_pinned_buffer(self, nbytes: int) -> memoryview:
         """ReadReq.buf_alloc hook: the storage layer reads directly into
@@ -390,6 +393,20 @@ class TensorBufferConsumer(BufferConsumer):
             return self.tensor_out.device
         return None
 
+    def will_verify_on_device(self) -> bool:
+        """True when consume_buffer will take the pinned->H2D device path
+        and can therefore checksum the bytes on-device at HBM speed
+        instead of CPU-hashing them in the read pipeline."""
+        from ..ops.staging import HIP_EXT_AVAILABLE
+
+        return (
+            HIP_EXT_AVAILABLE
+            and self._pinned_block is not None
+            and self.tensor_out is not None
+            and self.tensor_out.device.type == "cuda"
+            and self.entry.serializer == SERIALIZER_BUFFER
+        )
+
     def consume_from_device_u8(self, dev_u8: torch.Tensor) -> None:
         dtype = str_to_dtype(self.entry.dtype)
         loaded = (
@@ -421,6 +438,12 @@ class TensorBufferConsumer(BufferConsumer):
                 dev_u8 = self._pinned_block.tensor[:n].to(
                     self.tensor_out.device, non_blocking=False
                 )
+                if self.expected_psum is not None:
+                    from ..ops.staging import verify_device_psum
+
+                    verify_device_psum(
+                        dev_u8, self.expected_psum, self.entry.location
+                    )
                 loaded = (
                     dev_u8.view(dtype).reshape(tuple(self.entry.shape))
                     if dtype != torch.uint8

@@ -490,6 +490,77 @@ static long long scatter_h2d(const std::vector<unsigned long long>& flat, int n,
                      producer_stream, device, /*gather=*/false, 0);
 }
 
+// ---------------------------------------------------------------------------
+// device-side psum64 of a flat contiguous buffer (restore verification at
+// HBM speed: CPU psum64 runs at a few GB/s per thread and would bottleneck
+// a 50 GB/s warm restore; this kernel reads the just-H2D'd bytes once).
+// byte0 = file byte offset of base[0]; must be 8-aligned.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(kBlockThreads) void psum_flat_kernel(
+    const char* base, unsigned long long nbytes, unsigned long long byte0,
+    unsigned long long* out) {
+  unsigned long long acc = 0;
+  unsigned long long tid =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  unsigned long long nthreads = (unsigned long long)gridDim.x * blockDim.x;
+  unsigned long long nwords = nbytes >> 3;
+  unsigned long long w0 = byte0 >> 3;
+  const unsigned long long* words =
+      reinterpret_cast<const unsigned long long*>(base);
+  for (unsigned long long w = tid; w < nwords; w += nthreads) {
+    acc += words[w] * psum_mult(w0 + w);
+  }
+  if (tid == 0) {
+    // tail (< 8 bytes): lane-shifted into its word, like the CPU verifier
+    for (unsigned long long b = nwords << 3; b < nbytes; ++b) {
+      unsigned long long v = (unsigned char)base[b];
+      acc += (v << (8 * ((byte0 + b) & 7))) * psum_mult((byte0 + b) >> 3);
+    }
+  }
+  __shared__ unsigned long long sh[kBlockThreads];
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int s = kBlockThreads / 2; s > 0; s >>= 1) {
+    if ((int)threadIdx.x < s) sh[threadIdx.x] += sh[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    atomicAdd(out, sh[0]);
+  }
+}
+
+static unsigned long long psum64_dev(uintptr_t ptr, unsigned long long nbytes,
+                                     unsigned long long byte0,
+                                     uintptr_t producer_stream, int device) {
+  if (byte0 % 8 != 0) {
+    throw std::runtime_error("psum64_dev: byte0 must be 8-aligned");
+  }
+  if (ptr % 8 != 0) {
+    throw std::runtime_error("psum64_dev: buffer must be 8-aligned");
+  }
+  HIP_CHECK(hipSetDevice(device));
+  DeviceCtx& ctx = get_ctx(device);
+  chain_after(ctx.h2d_stream, producer_stream, device);
+  unsigned long long* out = nullptr;
+  HIP_CHECK(
+      hipMallocAsync(reinterpret_cast<void**>(&out), 8, ctx.h2d_stream));
+  HIP_CHECK(hipMemsetAsync(out, 0, 8, ctx.h2d_stream));
+  // HBM-bound reduction: 2048 workgroups fill all 8 XCDs with slack
+  unsigned int grid = 2048;
+  if (nbytes < (1u << 22)) grid = 64;
+  hipLaunchKernelGGL(psum_flat_kernel, dim3(grid), dim3(kBlockThreads), 0,
+                     ctx.h2d_stream, reinterpret_cast<const char*>(ptr),
+                     nbytes, byte0, out);
+  HIP_CHECK(hipGetLastError());
+  unsigned long long host_out = 0;
+  HIP_CHECK(hipMemcpyAsync(&host_out, out, 8, hipMemcpyDeviceToHost,
+                           ctx.h2d_stream));
+  HIP_CHECK(hipFreeAsync(out, ctx.h2d_stream));
+  HIP_CHECK(hipStreamSynchronize(ctx.h2d_stream));
+  return host_out;
+}
+
 static void op_wait(long long handle) {
   hipEvent_t ev = nullptr;
   int device = 0;
@@ -589,6 +660,10 @@ PYBIND11_MODULE(_csnap, m) {
         py::arg("hash_out_ptr") = 0);
   m.def("scatter_h2d", &scatter_h2d,
         "copy pinned host bytes to device and scatter into strided tensors");
+  m.def("psum64_dev", &psum64_dev,
+        "psum64 of a flat contiguous device buffer (blocking)",
+        py::arg("ptr"), py::arg("nbytes"), py::arg("byte0"),
+        py::arg("producer_stream"), py::arg("device"));
   m.def("wait", &op_wait, "block until an op completes");
   m.def("query", &op_query, "poll an op");
   m.def("is_managed_ptr", &is_managed_ptr);

@@ -562,6 +562,43 @@ def copy_buffer_via_pinned(
     return dev_u8.view(dtype).reshape(tuple(shape))
 
 
+def device_psum64(dev_u8: torch.Tensor, file_byte_off: int = 0) -> int:
+    """psum64 of a contiguous uint8 device tensor, computed on-device at
+    HBM speed (restore-verification path; ~free vs the H2D copy that
+    produced the bytes). ``file_byte_off`` is the byte offset of the
+    buffer's first byte within its payload file (8-aligned)."""
+    _require_ext()
+    assert dev_u8.dtype == torch.uint8 and dev_u8.is_contiguous()
+    assert dev_u8.device.type == "cuda"
+    if dev_u8.numel() == 0:
+        return 0
+    return int(
+        _csnap.psum64_dev(
+            dev_u8.data_ptr(),
+            dev_u8.numel(),
+            file_byte_off,
+            torch.cuda.current_stream(dev_u8.device).cuda_stream,
+            dev_u8.device.index,
+        )
+    )
+
+
+def verify_device_psum(
+    dev_u8: torch.Tensor, expected: "tuple[int, int]", what: str
+) -> None:
+    """Check a device buffer against (expected_psum_value, word_base);
+    raises like the CPU verifier on mismatch."""
+    value, word_base = expected
+    got = device_psum64(dev_u8, word_base * 8)
+    if got != value:
+        raise RuntimeError(
+            f"checksum mismatch for '{what}': snapshot recorded "
+            f"psum64:{value:016x}, device read back psum64:{got:016x} — "
+            "the file is corrupted or was modified after the snapshot "
+            "was committed"
+        )
+
+
 _engines: Dict[int, StagingEngine] = {}
 _engines_lock = threading.Lock()
 

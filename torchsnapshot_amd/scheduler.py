@@ -397,7 +397,29 @@ def execute_read_reqs(
                 stats.io_bytes += memoryview(buf).nbytes
                 if checksums:
                     loop = asyncio.get_running_loop()
-                    if req.byte_range is None:
+                    # device-bound consumers verify the bytes ON the GPU
+                    # right after their H2D copy (HBM-speed hash; a CPU
+                    # hash here would bottleneck a warm 50 GB/s restore)
+                    dev_exp = None
+                    try:
+                        if req.consumer.will_verify_on_device():
+                            if req.byte_range is None:
+                                want = checksums.get(req.path, "")
+                                if want.startswith("psum64:"):
+                                    dev_exp = (int(want[7:], 16), 0)
+                            else:
+                                s0, e0 = req.byte_range
+                                want = checksums.get(
+                                    integrity.member_key(req.path, s0, e0),
+                                    "",
+                                )
+                                if want.startswith("psum64:") and s0 % 8 == 0:
+                                    dev_exp = (int(want[7:], 16), s0 // 8)
+                    except AttributeError:
+                        pass
+                    if dev_exp is not None:
+                        req.consumer.expected_psum = dev_exp
+                    elif req.byte_range is None:
                         await loop.run_in_executor(
                             executor, integrity.verify_buffer, req.path, buf,
                             checksums,
