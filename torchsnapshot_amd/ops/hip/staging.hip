@@ -540,24 +540,26 @@ static unsigned long long psum64_dev(uintptr_t ptr, unsigned long long nbytes,
     throw std::runtime_error("psum64_dev: buffer must be 8-aligned");
   }
   HIP_CHECK(hipSetDevice(device));
-  DeviceCtx& ctx = get_ctx(device);
-  chain_after(ctx.h2d_stream, producer_stream, device);
+  // run on the CALLER's stream: the H2D copy that produced the bytes ran
+  // there (and completed), so the only added cost is the ~0.1 ms HBM-rate
+  // reduction itself. Routing through the shared side stream instead was
+  // measured at 7x restore slowdown (cross-thread chain_after + full
+  // side-stream syncs convoyed every consumer).
+  hipStream_t stream = reinterpret_cast<hipStream_t>(producer_stream);
   unsigned long long* out = nullptr;
-  HIP_CHECK(
-      hipMallocAsync(reinterpret_cast<void**>(&out), 8, ctx.h2d_stream));
-  HIP_CHECK(hipMemsetAsync(out, 0, 8, ctx.h2d_stream));
+  HIP_CHECK(hipMallocAsync(reinterpret_cast<void**>(&out), 8, stream));
+  HIP_CHECK(hipMemsetAsync(out, 0, 8, stream));
   // HBM-bound reduction: 2048 workgroups fill all 8 XCDs with slack
   unsigned int grid = 2048;
   if (nbytes < (1u << 22)) grid = 64;
   hipLaunchKernelGGL(psum_flat_kernel, dim3(grid), dim3(kBlockThreads), 0,
-                     ctx.h2d_stream, reinterpret_cast<const char*>(ptr),
-                     nbytes, byte0, out);
+                     stream, reinterpret_cast<const char*>(ptr), nbytes,
+                     byte0, out);
   HIP_CHECK(hipGetLastError());
   unsigned long long host_out = 0;
-  HIP_CHECK(hipMemcpyAsync(&host_out, out, 8, hipMemcpyDeviceToHost,
-                           ctx.h2d_stream));
-  HIP_CHECK(hipFreeAsync(out, ctx.h2d_stream));
-  HIP_CHECK(hipStreamSynchronize(ctx.h2d_stream));
+  HIP_CHECK(hipMemcpyAsync(&host_out, out, 8, hipMemcpyDeviceToHost, stream));
+  HIP_CHECK(hipFreeAsync(out, stream));
+  HIP_CHECK(hipStreamSynchronize(stream));
   return host_out;
 }
 
