@@ -248,6 +248,7 @@ __global__ __launch_bounds__(kBlockThreads) void pack_kernel(
 struct DeviceCtx {
   hipStream_t d2h_stream = nullptr;
   hipStream_t h2d_stream = nullptr;
+  hipStream_t verify_stream = nullptr;  // lazily created (psum64_dev)
 };
 
 struct Op {
@@ -540,12 +541,23 @@ static unsigned long long psum64_dev(uintptr_t ptr, unsigned long long nbytes,
     throw std::runtime_error("psum64_dev: buffer must be 8-aligned");
   }
   HIP_CHECK(hipSetDevice(device));
-  // run on the CALLER's stream: the H2D copy that produced the bytes ran
-  // there (and completed), so the only added cost is the ~0.1 ms HBM-rate
-  // reduction itself. Routing through the shared side stream instead was
-  // measured at 7x restore slowdown (cross-thread chain_after + full
-  // side-stream syncs convoyed every consumer).
-  hipStream_t stream = reinterpret_cast<hipStream_t>(producer_stream);
+  // The caller guarantees the bytes are already materialized (the H2D
+  // copy that produced them was synchronous), so NO dependency on any
+  // other stream is needed. A dedicated non-blocking stream keeps the
+  // final sync scoped to this ~0.1 ms reduction alone — syncing the
+  // caller's (legacy default) stream instead convoyed on every other
+  // consumer thread's queued work and cost a 13x restore slowdown.
+  (void)producer_stream;
+  hipStream_t stream;
+  {
+    DeviceCtx& ctx = get_ctx(device);
+    std::lock_guard<std::mutex> lk(g_mu);
+    if (ctx.verify_stream == nullptr) {
+      HIP_CHECK(hipStreamCreateWithFlags(&ctx.verify_stream,
+                                         hipStreamNonBlocking));
+    }
+    stream = ctx.verify_stream;
+  }
   unsigned long long* out = nullptr;
   HIP_CHECK(hipMallocAsync(reinterpret_cast<void**>(&out), 8, stream));
   HIP_CHECK(hipMemsetAsync(out, 0, 8, stream));
