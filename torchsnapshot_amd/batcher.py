@@ -260,6 +260,9 @@ class BatchedBufferConsumer(BufferConsumer):
         self.span_start = span_start
         self._pinned_block = None
         self._pinned_nbytes = 0
+        # (expected psum64 of the whole span, word_base) set by the read
+        # scheduler; verified on-device right after the span's single H2D
+        self.expected_psum = None
 
     def alloc_pinned_buffer(self, nbytes: int) -> memoryview:
         self._pinned_block = get_pinned_pool().acquire(max(nbytes, 1))
@@ -296,6 +299,11 @@ class BatchedBufferConsumer(BufferConsumer):
                 return None
         return device
 
+    def will_verify_on_device(self) -> bool:
+        from .ops.staging import HIP_EXT_AVAILABLE
+
+        return HIP_EXT_AVAILABLE and self._device_fast_path_target() is not None
+
     async def consume_buffer(self, ctx: StageContext, buf: BufferType) -> None:
         device = self._device_fast_path_target()
         if device is None:
@@ -326,6 +334,15 @@ class BatchedBufferConsumer(BufferConsumer):
                     )
                 finally:
                     pool.release(block)
+            if self.expected_psum is not None:
+                from .ops.staging import verify_device_psum
+
+                verify_device_psum(
+                    dev_span,
+                    self.expected_psum,
+                    f"{self.members[0].path} span[{self.span_start}:"
+                    f"{self.span_start + memoryview(buf).nbytes}]",
+                )
             for m in self.members:
                 s, e = m.byte_range
                 sub = dev_span[s - self.span_start : e - self.span_start]

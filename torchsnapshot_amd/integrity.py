@@ -139,19 +139,13 @@ def psum64_value(buf, word_base: int = 0) -> int:
     return int(psum64_hexdigest(buf, word_base)[len("psum64:"):], 16)
 
 
-def verify_ranged_buffer(
-    path: str, buf, byte_range, expected: Dict[str, str]
-) -> bool:
-    """Verify a byte-range read (a batched-slab member or a merged span of
-    members) against per-member psum64 values recorded at save time.
-    Returns True when verification happened, False when the range had no
-    recorded members (the caller may then fall back to whole-file
-    partial-sum accumulation); raises on mismatch.
-
-    psum64 is additive over disjoint file word-ranges and slab padding is
-    zeroed, so the expected checksum of any member-aligned span is the
-    (mod 2^64) sum of the recorded per-member values inside it.
-    """
+def expected_span_psum(
+    path: str, byte_range, expected: Dict[str, str]
+) -> Optional[int]:
+    """Expected psum64 of the file span ``byte_range`` of ``path``: the
+    (mod 2^64) sum of every recorded member checksum inside it (slab
+    padding between members is zeroed and contributes nothing). None when
+    no member of the span was recorded."""
     start, end = byte_range
     prefix = path + "#"
     total = 0
@@ -166,7 +160,25 @@ def verify_ranged_buffer(
         if s >= start and e <= end:
             total = (total + int(v[len("psum64:"):], 16)) % (1 << 64)
             found = True
-    if not found:
+    return total if found else None
+
+
+def verify_ranged_buffer(
+    path: str, buf, byte_range, expected: Dict[str, str]
+) -> bool:
+    """Verify a byte-range read (a batched-slab member or a merged span of
+    members) against per-member psum64 values recorded at save time.
+    Returns True when verification happened, False when the range had no
+    recorded members (the caller may then fall back to whole-file
+    partial-sum accumulation); raises on mismatch.
+
+    psum64 is additive over disjoint file word-ranges and slab padding is
+    zeroed, so the expected checksum of any member-aligned span is the
+    (mod 2^64) sum of the recorded per-member values inside it.
+    """
+    start, end = byte_range
+    total = expected_span_psum(path, byte_range, expected)
+    if total is None:
         return False
     if start % 8 != 0:
         # slab members are 64-byte aligned; a misaligned span can't use

@@ -409,12 +409,11 @@ def execute_read_reqs(
                                     dev_exp = (int(want[7:], 16), 0)
                             else:
                                 s0, e0 = req.byte_range
-                                want = checksums.get(
-                                    integrity.member_key(req.path, s0, e0),
-                                    "",
+                                total = integrity.expected_span_psum(
+                                    req.path, req.byte_range, checksums
                                 )
-                                if want.startswith("psum64:") and s0 % 8 == 0:
-                                    dev_exp = (int(want[7:], 16), s0 // 8)
+                                if total is not None and s0 % 8 == 0:
+                                    dev_exp = (total, s0 // 8)
                     except AttributeError:
                         pass
                     if dev_exp is not None:
