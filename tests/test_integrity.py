@@ -176,3 +176,25 @@ def test_tiled_read_verification(monkeypatch):
             f.write(bytes([b[0] ^ 0x1]))
         with pytest.raises(RuntimeError, match="tiled read"):
             snap.read_object("0/sd/big", memory_budget_bytes=64 * 1024)
+
+
+def test_stale_checksum_file_cleared_on_unchecksummed_resave(monkeypatch):
+    """Re-saving WITHOUT checksums to a path that previously held a
+    checksummed snapshot must clear the old .checksums file — otherwise
+    verification compares new payloads against stale values and reports
+    phantom corruption."""
+    sd = StateDict(a=torch.rand(256, 16))
+    with tmp_snapshot_path() as path:
+        monkeypatch.setenv("TSAMD_CHECKSUM", "1")
+        Snapshot.take(path, {"sd": sd})
+        assert os.path.exists(os.path.join(path, "0", ".checksums"))
+
+        monkeypatch.setenv("TSAMD_CHECKSUM", "0")
+        sd2 = StateDict(a=torch.rand(256, 16))  # different content
+        snap = Snapshot.take(path, {"sd": sd2})
+        assert not os.path.exists(os.path.join(path, "0", ".checksums"))
+
+        monkeypatch.setenv("TSAMD_VERIFY_CHECKSUM", "1")
+        out = StateDict()
+        snap.restore({"sd": out})  # no phantom mismatch
+        assert torch.equal(out["a"], sd2["a"])

@@ -95,6 +95,28 @@ def write_checksum_file(
     storage: StoragePlugin, rank: int, checksums: Dict[str, str]
 ) -> None:
     if not checksums:
+        # an unchecksummed save to a path that previously held a
+        # checksummed snapshot must not leave the old rank file behind —
+        # verification would compare the NEW payloads against the stale
+        # values and report phantom corruption
+        from .scheduler import run_coro_sync
+
+        async def cleanup() -> None:
+            try:
+                await storage.delete(checksum_file_path(rank))
+            except (FileNotFoundError, OSError):
+                pass
+            except Exception:
+                logger.debug(
+                    "could not remove stale checksum file", exc_info=True
+                )
+            finally:
+                try:
+                    await storage.close_for_loop()
+                except Exception:
+                    pass
+
+        run_coro_sync(cleanup())
         return
     storage.sync_write(
         WriteIO(
