@@ -461,3 +461,53 @@ def test_device_verified_restore_roundtrip(monkeypatch):
                     )
                 }
             )
+
+
+def test_kitchen_sink_gpu_everything_on(monkeypatch):
+    """Maximal integration: chunking + batching + checksums + async take +
+    device-verified restore + tiled read_object, all in one snapshot."""
+    monkeypatch.setenv("TSAMD_CHECKSUM", "1")
+    monkeypatch.setenv("TSAMD_MAX_CHUNK_SIZE_BYTES", str(2 * 1024 * 1024))
+    torch.manual_seed(7)
+    sd = StateDict(
+        big=torch.randn(2048, 1024, device="cuda"),  # 8 MB -> 4 chunks
+        **{f"s{i}": torch.randn(199, 33, device="cuda") for i in range(9)},
+        host=torch.randn(321, 5),
+        t_view=torch.randn(128, 128, device="cuda").t(),
+        n=41,
+        f=2.5,
+    )
+    saved = {
+        k: (v.clone() if isinstance(v, torch.Tensor) else v)
+        for k, v in sd.items()
+    }
+    with tmp_snapshot_path() as path:
+        pending = Snapshot.async_take(path, {"sd": sd})
+        # mutate everything while I/O is in flight
+        with torch.no_grad():
+            for v in sd.values():
+                if isinstance(v, torch.Tensor):
+                    v.zero_()
+        snap = pending.wait()
+
+        monkeypatch.setenv("TSAMD_VERIFY_CHECKSUM", "1")
+        out = StateDict(
+            big=torch.zeros(2048, 1024, device="cuda"),
+            **{f"s{i}": torch.zeros(199, 33, device="cuda") for i in range(9)},
+            host=torch.zeros(321, 5),
+            t_view=torch.zeros(128, 128, device="cuda"),
+            n=0,
+            f=0.0,
+        )
+        snap.restore({"sd": out})
+        for k, v in saved.items():
+            if isinstance(v, torch.Tensor):
+                got = out[k]
+                want = v.contiguous().to(got.device)
+                assert torch.equal(got, want), k
+            else:
+                assert out[k] == v, k
+
+        # tiled random access with verification active
+        big = snap.read_object("0/sd/big", memory_budget_bytes=1024 * 1024)
+        assert torch.equal(big.cpu(), saved["big"].cpu())
