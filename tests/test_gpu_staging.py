@@ -42,6 +42,8 @@ def test_ext_available():
         ((128,), torch.float64, "contig"),
         ((0, 4), torch.float32, "contig"),
         ((77, 3), torch.uint8, "contig"),
+        ((4096, 512), torch.float8_e4m3fn, "contig"),
+        ((255, 17), torch.float8_e5m2, "contig"),
     ],
 )
 def test_stage_single(shape, dtype, make):
@@ -127,6 +129,7 @@ def test_scatter_h2d_round_trip():
 
 def test_snapshot_gpu_round_trip():
     sd = StateDict(
+        fp8=torch.randn(512, 256).to(torch.float8_e4m3fn).cuda(),
         big=torch.randn(2048, 2048, dtype=torch.bfloat16, device="cuda"),
         small1=torch.randn(100, device="cuda"),
         small2=torch.randn(64, 64, device="cuda").t(),
@@ -136,6 +139,7 @@ def test_snapshot_gpu_round_trip():
     with tmp_snapshot_path() as path:
         snap = Snapshot.take(path, {"sd": sd})
         out = StateDict(
+            fp8=torch.zeros(512, 256, dtype=torch.float8_e4m3fn, device="cuda"),
             big=torch.zeros(2048, 2048, dtype=torch.bfloat16, device="cuda"),
             small1=torch.zeros(100, device="cuda"),
             small2=torch.zeros(64, 64, device="cuda"),
@@ -143,6 +147,9 @@ def test_snapshot_gpu_round_trip():
             n=0,
         )
         snap.restore({"sd": out})
+        assert torch.equal(
+            out["fp8"].view(torch.uint8), sd["fp8"].view(torch.uint8)
+        )
         assert torch.equal(out["big"], sd["big"])
         assert torch.equal(out["small1"], sd["small1"])
         assert torch.equal(out["small2"], sd["small2"].contiguous())
