@@ -192,3 +192,71 @@ def test_async_take_custom_tensor_prepare_func():
         out = StateDict(w=torch.zeros(64, 32))
         snap.restore({"sd": out})
         assert torch.equal(out["w"], sd["w"].to(torch.bfloat16).float())
+
+
+def test_async_shadow_returns_without_staging_wait(monkeypatch):
+    """With shadow clones, async_take must not block on staging: inject a
+    slow stager-side storage and check the return races ahead of I/O
+    while mutation safety still holds."""
+    monkeypatch.setenv("TSAMD_ASYNC_SHADOW", "1")
+    sd = StateDict(w=torch.rand(256, 256), b=torch.rand(64))
+    saved_w = sd["w"].clone()
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "snap")
+        with _patch_plugin(SlowFSStoragePlugin):
+            pending = Snapshot.async_take(path, {"sd": sd})
+            assert pending.sources_immutable
+            sd["w"].fill_(-123.0)  # mutate immediately
+            snap = pending.wait()
+        out = StateDict()
+        snap.restore({"sd": out})
+        assert torch.equal(out["w"], saved_w)
+
+
+def test_async_shadow_falls_back_on_unshadowable_leaves(monkeypatch):
+    """Objects that can't be cloned safely force the classic
+    wait-for-staging path (sources_immutable False)."""
+    monkeypatch.setenv("TSAMD_ASYNC_SHADOW", "1")
+    # a set is a picklable non-tensor LEAF (flatten doesn't descend into
+    # it), so it lands in the object preparer and can't be shadowed
+    sd = StateDict(w=torch.rand(16), blob={1, 2, 3})
+    with tempfile.TemporaryDirectory() as d:
+        pending = Snapshot.async_take(os.path.join(d, "snap"), {"sd": sd})
+        assert not pending.sources_immutable
+        pending.wait()
+
+
+def test_async_shadow_tied_weights_still_dedup(monkeypatch):
+    monkeypatch.setenv("TSAMD_ASYNC_SHADOW", "1")
+    shared = torch.rand(128)
+
+    class Tied:
+        def state_dict(self):
+            return {"a": shared, "b": shared}
+
+        def load_state_dict(self, sd):
+            pass
+
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "snap")
+        pending = Snapshot.async_take(path, {"m": Tied()})
+        snap = pending.wait()
+        man = snap.get_manifest()
+        # both paths point at ONE payload (same location + byte range)
+        a, b = man["0/m/a"], man["0/m/b"]
+        assert (a.get("location"), a.get("byte_range")) == (
+            b.get("location"),
+            b.get("byte_range"),
+        )
+        out = {}
+
+        class Out:
+            def state_dict(self):
+                return {"a": torch.zeros(128), "b": torch.zeros(128)}
+
+            def load_state_dict(self, sd):
+                out.update(sd)
+
+        snap.restore({"m": Out()})
+        assert torch.equal(out["a"], shared)
+        assert torch.equal(out["b"], shared)

@@ -192,6 +192,7 @@ class BatchedBufferStager(BufferStager):
         if self._staged_batch is not None:
             self._staged_batch.release()
             self._staged_batch = None
+        self.tensors = ()  # free shadow clones as soon as the write lands
 
 
 # ---------------------------------------------------------------------------

@@ -342,6 +342,10 @@ class TensorBufferStager(BufferStager):
         if self._staged_batch is not None:
             self._staged_batch.release()
             self._staged_batch = None
+        # drop the source reference: for shadow-cloned async saves this
+        # frees the device clone as soon as its write completes (any
+        # zero-copy host buffer stays alive through the memoryview)
+        self.tensor = None  # type: ignore[assignment]
 
 
 class TensorBufferConsumer(BufferConsumer):

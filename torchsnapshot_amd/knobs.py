@@ -114,6 +114,22 @@ def is_hip_staging_disabled() -> bool:
     return _env_flag("TSAMD_DISABLE_HIP_STAGING")
 
 
+def get_async_shadow_mode() -> str:
+    """Shadow-clone async snapshots: "auto" (default) clones device
+    tensors at HBM rate before async_take returns, when free HBM allows —
+    training resumes after milliseconds instead of waiting for the D2H
+    staging of the whole model ("1" forces it, "0" disables)."""
+    val = os.environ.get("TSAMD_ASYNC_SHADOW", "auto").lower()
+    return val if val in ("auto", "1", "0") else "auto"
+
+
+def get_shadow_cpu_max_bytes() -> int:
+    """Cap on CPU-tensor bytes eagerly cloned in shadow mode: large host
+    states clone faster in the parallel staging pipeline than on the
+    caller thread, so beyond this the classic wait-for-staging path wins."""
+    return _env_bytes("TSAMD_SHADOW_CPU_MAX_BYTES", 1024 * _MB)
+
+
 def get_storage_write_chunk_bytes() -> int:
     """Chunk size for filesystem pwrite calls (large sequential writes)."""
     return _env_bytes("TSAMD_FS_WRITE_CHUNK_BYTES", 256 * _MB)

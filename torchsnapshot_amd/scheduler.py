@@ -139,6 +139,10 @@ class PendingIOWork:
         self._exc_holder = exc_holder
         # {payload_path: xxh3 hex}, filled when TSAMD_CHECKSUM=1
         self.checksums = checksums if checksums is not None else {}
+        # True when every source was shadow-cloned before the pipeline
+        # started (async saves): the app may mutate its state without
+        # waiting for staging
+        self.sources_immutable = False
 
     def _maybe_raise(self) -> None:
         if self._exc_holder:

@@ -264,8 +264,11 @@ class Snapshot:
                 is_async=True,
                 custom_tensor_prepare_func=_custom_tensor_prepare_func,
             )
-            # once staging is complete, the app may mutate its state freely
-            pending_io_work.wait_staged()
+            # once staging is complete, the app may mutate its state
+            # freely. With shadow clones (sources_immutable) there is
+            # nothing to wait for — the pipeline owns private copies.
+            if not pending_io_work.sources_immutable:
+                pending_io_work.wait_staged()
         except Exception as e:
             # peers are (or will be) waiting in the commit barrier: tell
             # them this rank failed before they time out
@@ -338,6 +341,15 @@ class Snapshot:
                 if isinstance(obj, torch.Tensor):
                     flattened[p] = custom_tensor_prepare_func(p, obj)
 
+        # zero-stall async saves: shadow-clone the state before staging so
+        # training may resume immediately (device clones run at HBM rate —
+        # milliseconds for a whole model on 288 GB HBM3E; vs waiting for
+        # the full D2H staging pass). Cloning preserves aliasing, so
+        # tied-weight dedup below still sees shared storage.
+        sources_immutable = False
+        if is_async:
+            sources_immutable = cls._shadow_for_async(flattened)
+
         write_reqs: List[WriteReq] = []
         req_to_logical: Dict[str, str] = {}
         # tied-weight dedup: identical tensor objects (same storage, view,
@@ -397,7 +409,89 @@ class Snapshot:
         pending = execute_write_reqs(
             write_reqs, storage, budget, rank=rank, is_async=is_async
         )
+        pending.sources_immutable = sources_immutable
         return pending, metadata
+
+    @classmethod
+    def _shadow_for_async(cls, flattened: Flattened) -> bool:
+        """Replace every tensor leaf with a private clone (device clones
+        at HBM rate; identical tensors share ONE clone so tied-weight
+        dedup survives). Returns True iff EVERY leaf is now immutable-safe
+        — then async_take may return without waiting for staging. Falls
+        back (False) when a leaf can't be shadowed (ShardedTensor,
+        arbitrary objects), device memory is too tight, or host-tensor
+        bytes exceed the eager-clone budget."""
+        mode = knobs.get_async_shadow_mode()
+        if mode == "0":
+            return False
+        try:
+            from torch.distributed.tensor import DTensor
+        except ImportError:
+            DTensor = ()  # type: ignore[assignment]
+
+        tensor_paths: List[str] = []
+        cuda_need: Dict[int, int] = {}
+        cpu_bytes = 0
+        for p, obj in flattened.items():
+            if isinstance(obj, (int, float, str, bool, bytes)) or obj is None:
+                continue
+            if DTensor and isinstance(obj, DTensor):
+                local = obj.to_local()
+                nbytes = local.numel() * local.element_size()
+                if local.device.type == "cuda":
+                    idx = local.device.index or 0
+                    cuda_need[idx] = cuda_need.get(idx, 0) + nbytes
+                else:
+                    cpu_bytes += nbytes
+                tensor_paths.append(p)
+            elif isinstance(obj, torch.Tensor) and (
+                type(obj) is torch.Tensor
+                or isinstance(obj, torch.nn.Parameter)
+            ):
+                nbytes = obj.numel() * obj.element_size()
+                if obj.device.type == "cuda":
+                    idx = obj.device.index or 0
+                    cuda_need[idx] = cuda_need.get(idx, 0) + nbytes
+                else:
+                    cpu_bytes += nbytes
+                tensor_paths.append(p)
+            else:
+                # ShardedTensor / quantized / arbitrary object: staging
+                # must complete before the app may mutate it
+                return False
+        if cpu_bytes > knobs.get_shadow_cpu_max_bytes():
+            return False
+        if mode == "auto":
+            for idx, need in cuda_need.items():
+                try:
+                    free, _total = torch.cuda.mem_get_info(idx)
+                except Exception:
+                    return False
+                if free < int(need * 1.25):
+                    logger.info(
+                        "async shadow disabled: device %d has %.1f GB free "
+                        "but the shadow needs %.1f GB",
+                        idx, free / 1e9, need / 1e9,
+                    )
+                    return False
+        # clone, preserving aliasing (tied weights -> one shared clone)
+        clones: Dict[Any, Any] = {}
+        for p in tensor_paths:
+            obj = flattened[p]
+            local = obj.to_local() if (DTensor and isinstance(obj, DTensor)) else obj
+            key = (
+                local.data_ptr(),
+                local.dtype,
+                tuple(local.shape),
+                tuple(local.stride()),
+                str(local.device),
+            )
+            if key not in clones:
+                clones[key] = obj.detach().clone()
+            flattened[p] = clones[key]
+        for idx in cuda_need:
+            torch.cuda.synchronize(idx)
+        return True
 
     @classmethod
     def _partition_replicated(
@@ -875,6 +969,9 @@ class PendingSnapshot:
     ) -> None:
         self.path = path
         self._pending_io_work = pending_io_work
+        # True when the snapshot's sources were shadow-cloned: the app
+        # was free to mutate its state the moment async_take returned
+        self.sources_immutable = pending_io_work.sources_immutable
         self._pg_wrapper = pg_wrapper
         self._metadata = metadata
         self._storage = storage
