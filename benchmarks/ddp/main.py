@@ -31,8 +31,11 @@ from torchsnapshot_amd.state_dict import StateDict
 
 def main() -> None:
     parser = argparse.ArgumentParser()
+    # param-size is in BYTES, matching the reference exactly (its model
+    # builds torch.rand(param_size / 4) f32 elements per parameter,
+    # reference benchmarks/ddp/main.py:25,38): 100 MB x 200 = 20 GB
     parser.add_argument("--param-size", type=int, default=int(1e8))
-    parser.add_argument("--num-params", type=int, default=200)  # 20 GB
+    parser.add_argument("--num-params", type=int, default=200)
     parser.add_argument("--work-dir", default="/tmp/tsamd_ddp_bench")
     parser.add_argument("--device", default="cuda")
     parser.add_argument("--compare-torch-save", action="store_true")
@@ -54,7 +57,9 @@ def main() -> None:
     sd = StateDict()
     total_bytes = 0
     for i in range(args.num_params):
-        t = torch.empty(args.param_size, dtype=torch.float32, device=device)
+        t = torch.empty(
+            args.param_size // 4, dtype=torch.float32, device=device
+        )
         t.uniform_(-1, 1)
         sd[f"param_{i}"] = t
         total_bytes += t.numel() * 4
