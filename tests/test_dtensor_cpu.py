@@ -230,3 +230,37 @@ def test_dtensor_uneven_save3_restore5():
     with tempfile.TemporaryDirectory() as d:
         run_multiprocess(3, _save_uneven, d)
         run_multiprocess(5, _restore_uneven, d)
+
+
+def _async_shadow_dtensor(tmpdir: str) -> None:
+    import os as _os
+
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import distribute_tensor
+    from torch.distributed.tensor.placement_types import Shard
+
+    from torchsnapshot_amd import Snapshot
+
+    _os.environ["TSAMD_ASYNC_SHADOW"] = "1"
+    try:
+        mesh = init_device_mesh("cpu", (dist.get_world_size(),))
+        dt = distribute_tensor(_full(31), mesh, [Shard(0)])
+        holder = _Holder(dt)
+        path = _os.path.join(tmpdir, "snap")
+        pending = Snapshot.async_take(path, {"obj": holder})
+        assert pending.sources_immutable
+        # mutate the local shard immediately — the snapshot must not see it
+        with torch.no_grad():
+            dt.to_local().zero_()
+        pending.wait()
+
+        out = _Holder(distribute_tensor(torch.zeros(48, 8), mesh, [Shard(0)]))
+        Snapshot(path).restore({"obj": out})
+        assert torch.equal(out.dt.full_tensor(), _full(31))
+    finally:
+        del _os.environ["TSAMD_ASYNC_SHADOW"]
+
+
+def test_async_shadow_dtensor_world2():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(2, _async_shadow_dtensor, d)
