@@ -64,8 +64,10 @@ def is_partitioner_disabled() -> bool:
 # -- execution ---------------------------------------------------------------
 
 def get_max_io_concurrency() -> int:
-    """Maximum concurrent storage I/O operations per rank."""
-    return _env_int("TSAMD_MAX_PER_RANK_IO_CONCURRENCY", 24)
+    """Maximum concurrent storage I/O operations per rank (32 measured
+    best for both the steady-state save and the cold parallel-segment
+    restore on MI355X NVMe, profiles/r02_measurements.md)."""
+    return _env_int("TSAMD_MAX_PER_RANK_IO_CONCURRENCY", 32)
 
 
 def get_num_staging_threads() -> int:
