@@ -413,3 +413,55 @@ def test_replicated_dtensor_device_pieces_world2_shared_gpu():
             backend="gloo",
             share_device=True,
         )
+
+
+def _world4_mixed_shared(tmpdir: str) -> None:
+    rank = dist.get_rank()
+    torch.manual_seed(3)  # identical replicated content everywhere
+    sd = StateDict(
+        shared=torch.rand(256, 64, device="cuda"),
+        mine=torch.full((100,), float(rank), device="cuda"),
+        **{f"w{i}": torch.rand(64, 16, device="cuda") for i in range(4)},
+    )
+    saved = {k: v.clone() for k, v in sd.items()}
+    path = os.path.join(tmpdir, "snap")
+    Snapshot.take(path, {"sd": sd}, replicated=["sd/shared"])
+    out = StateDict(
+        shared=torch.zeros(256, 64, device="cuda"),
+        mine=torch.zeros(100, device="cuda"),
+        **{f"w{i}": torch.zeros(64, 16, device="cuda") for i in range(4)},
+    )
+    Snapshot(path).restore({"sd": out})
+    for k, v in saved.items():
+        assert torch.equal(out[k], v), k
+
+
+def test_world4_partitioner_shared_gpu():
+    """Four ranks on one GPU: replicated partitioning + per-rank batching
+    at world 4 (the 8-GPU node's layout, minus RCCL)."""
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess_gpu(
+            4, _world4_mixed_shared, d, backend="gloo", share_device=True
+        )
+
+
+def test_async_shadow_no_device_leak():
+    """Back-to-back shadowed async saves must return all device memory:
+    the clones are freed as their writes land."""
+    import gc
+
+    sd = StateDict(w=torch.randn(1024, 1024, device="cuda"))  # 4 MB
+    with tempfile.TemporaryDirectory() as d:
+        # warm up allocator + pools
+        p0 = Snapshot.async_take(os.path.join(d, "s_warm"), {"sd": sd})
+        p0.wait()
+        gc.collect()
+        torch.cuda.empty_cache()
+        base = torch.cuda.memory_allocated()
+        for i in range(8):
+            pending = Snapshot.async_take(os.path.join(d, f"s{i}"), {"sd": sd})
+            assert pending.sources_immutable
+            pending.wait()
+        gc.collect()
+        grown = torch.cuda.memory_allocated() - base
+        assert grown < 8 * 1024 * 1024, f"device memory grew {grown} bytes"
