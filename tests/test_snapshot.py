@@ -416,3 +416,25 @@ def test_parallel_segment_fs_io_fsync(monkeypatch):
         out = StateDict()
         snap.restore({"sd": out})
         assert torch.equal(out["big"], sd["big"])
+
+
+def test_many_snapshots_loop_stability():
+    """50 back-to-back take/async_take/restore cycles: no pipeline-thread,
+    event-loop, or store-state leakage across snapshots."""
+    import torch
+
+    from torchsnapshot_amd import Snapshot, StateDict
+    from torchsnapshot_amd.test_utils import tmp_snapshot_path
+
+    sd = StateDict(w=torch.rand(64, 64), step=0)
+    with tmp_snapshot_path() as path:
+        for i in range(50):
+            sd["step"] = i
+            if i % 2:
+                snap = Snapshot.async_take(path, {"sd": sd}).wait()
+            else:
+                snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict()
+        snap.restore({"sd": out})
+        assert out["step"] == 49
+        assert torch.equal(out["w"], sd["w"])
