@@ -148,3 +148,41 @@ def _async_take_same_path_twice(tmpdir: str) -> None:
 def test_async_take_same_path_reuse_world2():
     with tempfile.TemporaryDirectory() as d:
         run_multiprocess(2, _async_take_same_path_twice, d)
+
+
+# ---------------------------------------------------------------------------
+# shadow clones x replicated partitioner x tied aliases (all round-2
+# features composed, async, world 2)
+# ---------------------------------------------------------------------------
+
+
+def _tied_async_shadow(tmpdir: str) -> None:
+    from torchsnapshot_amd import Snapshot
+
+    os.environ["TSAMD_ASYNC_SHADOW"] = "1"
+    try:
+        path = os.path.join(tmpdir, "snap")
+        state = _tied_state()
+        pending = Snapshot.async_take(
+            path, {"m": state}, replicated=["m/a", "m/b"]
+        )
+        assert pending.sources_immutable
+        # mutate everything immediately
+        with torch.no_grad():
+            state.shared.zero_()
+            state.filler.zero_()
+        pending.wait()
+
+        target = _tied_state()
+        target.shared = torch.zeros(1024)
+        target.filler = torch.zeros_like(target.filler)
+        Snapshot(path).restore({"m": target})
+        torch.manual_seed(7)
+        assert torch.equal(target.shared, torch.rand(1024))
+    finally:
+        del os.environ["TSAMD_ASYNC_SHADOW"]
+
+
+def test_tied_replicated_async_shadow_world2():
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(2, _tied_async_shadow, d)
