@@ -432,6 +432,26 @@ class Snapshot:
         tensor_paths: List[str] = []
         cuda_need: Dict[int, int] = {}
         cpu_bytes = 0
+        try:
+            return cls._classify_and_shadow(
+                flattened, tensor_paths, cuda_need, cpu_bytes, mode, DTensor
+            )
+        except Exception:
+            # exotic leaves (sparse tensors raise on data_ptr, etc.):
+            # fall back to the classic wait-for-staging path
+            logger.debug("async shadow disabled by exception", exc_info=True)
+            return False
+
+    @classmethod
+    def _classify_and_shadow(
+        cls,
+        flattened: Flattened,
+        tensor_paths: List[str],
+        cuda_need: Dict[int, int],
+        cpu_bytes: int,
+        mode: str,
+        DTensor: Any,
+    ) -> bool:
         for p, obj in flattened.items():
             if isinstance(obj, (int, float, str, bool, bytes)) or obj is None:
                 continue
