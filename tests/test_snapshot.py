@@ -438,3 +438,29 @@ def test_many_snapshots_loop_stability():
         snap.restore({"sd": out})
         assert out["step"] == 49
         assert torch.equal(out["w"], sd["w"])
+
+
+def test_sparse_tensor_round_trip():
+    """Sparse tensors route through torch_save (no flat storage) and
+    survive save/restore — sync and async, with shadowing falling back
+    gracefully."""
+    import torch
+
+    from torchsnapshot_amd import Snapshot, StateDict
+    from torchsnapshot_amd.test_utils import tmp_snapshot_path
+
+    sp = torch.sparse_coo_tensor([[0, 2], [1, 0]], [1.5, -2.0], (3, 3))
+    sd = StateDict(dense=torch.rand(8), sp=sp)
+    with tmp_snapshot_path() as path:
+        snap = Snapshot.take(path, {"sd": sd})
+        out = StateDict()
+        snap.restore({"sd": out})
+        assert torch.equal(out["sp"].to_dense(), sp.to_dense())
+        assert torch.equal(out["dense"], sd["dense"])
+
+        pending = Snapshot.async_take(path + "2", {"sd": sd})
+        assert not pending.sources_immutable  # sparse forces classic path
+        snap2 = pending.wait()
+        out2 = StateDict()
+        snap2.restore({"sd": out2})
+        assert torch.equal(out2["sp"].to_dense(), sp.to_dense())

@@ -89,6 +89,9 @@ def pick_serializer(tensor: torch.Tensor) -> str:
         ):
             return SERIALIZER_QTENSOR
         return SERIALIZER_TORCH_SAVE
+    if tensor.layout != torch.strided:
+        # sparse (COO/CSR/...) tensors have no flat storage to serialize
+        return SERIALIZER_TORCH_SAVE
     if tensor.dtype in DTYPE_TO_STR:
         return SERIALIZER_BUFFER
     return SERIALIZER_TORCH_SAVE

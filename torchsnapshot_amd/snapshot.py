@@ -363,6 +363,7 @@ class Snapshot:
                 isinstance(obj, torch.Tensor)
                 and type(obj) is torch.Tensor
                 and not obj.is_quantized
+                and obj.layout == torch.strided  # sparse has no data_ptr
             ):
                 alias_key = (
                     obj.data_ptr(),
