@@ -465,3 +465,30 @@ def test_async_shadow_no_device_leak():
         gc.collect()
         grown = torch.cuda.memory_allocated() - base
         assert grown < 8 * 1024 * 1024, f"device memory grew {grown} bytes"
+
+
+def test_rccl_helpers_world1_smoke():
+    """Execute the RCCL test helpers at world 1 on one GPU so API typos
+    surface on ANY box rather than first on the driver's 8-GPU node (the
+    world-N asserts and mesh shapes still only run at their gated sizes).
+    """
+    import tempfile as _tf
+
+    for fn in (
+        _fsdp1_fullshard_roundtrip,
+        _fsdp2_fully_shard_roundtrip,
+        _save_then_restore_dtensor_world1,
+        _ddp_world1,
+    ):
+        with _tf.TemporaryDirectory() as d:
+            run_multiprocess_gpu(1, fn, d)
+
+
+def _save_then_restore_dtensor_world1(tmpdir: str) -> None:
+    _dtensor_save_n(tmpdir)
+    _dtensor_restore_n(tmpdir)
+
+
+def _ddp_world1(tmpdir: str) -> None:
+    _ddp_save(tmpdir)
+    _ddp_restore(tmpdir)
