@@ -10,6 +10,8 @@ mkdir -p gpurun_out
 : > "$LOG"
 
 cat > /tmp/pmc_driver.py <<'EOF'
+import os, sys
+sys.path.insert(0, os.environ["GRAFT_REPO_ROOT"])
 import torch
 from torchsnapshot_amd.ops import staging
 
