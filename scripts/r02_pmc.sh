@@ -1,4 +1,9 @@
 #!/bin/bash
+# NOTE (r02): PMC counter collection WEDGED on this workload (rocprofv3
+# --pmc + the staging engine's hipMallocAsync/multi-stream pattern); the
+# run was killed at the gpurun limit. Kernel evidence ships via
+# --kernel-trace stats instead (profiles/r02_bench_kernel_stats.txt).
+# Kept for reference; do not re-run without an inner timeout.
 # PMC counters for the two hand-written kernels (pack gather + psum64
 # verify): FETCH_SIZE/WRITE_SIZE give actual memory traffic so we can
 # show the gather reads each byte once (no waste) and the psum kernel is
