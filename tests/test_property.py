@@ -200,3 +200,25 @@ def test_psum64_zero_padding_invariant(pad, seed):
     rng = random.Random(seed)
     buf = bytes(rng.getrandbits(8) for _ in range(256))
     assert psum64_hexdigest(buf) == psum64_hexdigest(buf + b"\0" * pad)
+
+
+@pytest.mark.parametrize("seed", [11, 222, 3333, 44444, 555555])
+def test_fuzz_snapshot_seeded(seed):
+    """Deterministic slices of the randomized snapshot fuzzer
+    (scripts/fuzz_snapshot.py): random nested state dicts across dtypes,
+    views, aliases, knob combinations — take/restore must round-trip
+    bit-exactly."""
+    import importlib.util
+    import os as _os
+
+    spec = importlib.util.spec_from_file_location(
+        "fuzz_snapshot",
+        _os.path.join(
+            _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))),
+            "scripts",
+            "fuzz_snapshot.py",
+        ),
+    )
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.one_case(seed)
