@@ -202,6 +202,9 @@ def test_psum64_zero_padding_invariant(pad, seed):
     assert psum64_hexdigest(buf) == psum64_hexdigest(buf + b"\0" * pad)
 
 
+import pytest
+
+
 @pytest.mark.parametrize("seed", [11, 222, 3333, 44444, 555555])
 def test_fuzz_snapshot_seeded(seed):
     """Deterministic slices of the randomized snapshot fuzzer
