@@ -125,3 +125,22 @@ def test_qtensor_layout_round_trip():
     assert tensor_eq(t, t2)
     assert torch.equal(t2.q_per_channel_scales(), t.q_per_channel_scales())
     assert t2.q_per_channel_axis() == 0
+
+
+def test_memoryview_of_size1_weird_stride():
+    """A step-2 slice of a 2-element tensor has shape (1,), stride (2,):
+    is_contiguous() is True (size-1 dims allow any stride) but a uint8
+    reinterpret needs unit stride — found by the snapshot fuzzer
+    (seed 313949245)."""
+    import torch
+
+    from torchsnapshot_amd.serialization import (
+        tensor_as_memoryview,
+        tensor_from_memoryview,
+    )
+
+    t = torch.tensor([7, 9], dtype=torch.int16)[::2]
+    assert t.is_contiguous() and t.stride() == (2,)
+    mv = tensor_as_memoryview(t)
+    back = tensor_from_memoryview(mv, dtype=torch.int16, shape=(1,))
+    assert back.item() == 7

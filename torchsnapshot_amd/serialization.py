@@ -116,6 +116,12 @@ def tensor_as_memoryview(tensor: torch.Tensor) -> memoryview:
     flat = tensor.reshape(-1)
     if flat.numel() == 0:
         return memoryview(b"")
+    if flat.stride(0) != 1:
+        # size-1 dims make is_contiguous() true for ANY stride (e.g. a
+        # step-2 slice of a 2-element tensor -> shape (1,), stride (2,)),
+        # but the uint8 reinterpret needs a unit stride; re-materialize
+        # (trivial: such tensors have one element per weird dim)
+        flat = torch.as_strided(flat.clone(), flat.shape, (1,))
     u8 = flat.view(torch.uint8)
     return memoryview(u8.numpy())  # shares memory with the tensor
 
