@@ -178,7 +178,16 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=50)
     ap.add_argument("--seed", type=int, default=None)
+    ap.add_argument("--dist", action="store_true", help="world-2 gloo fuzz")
     args = ap.parse_args()
+    if args.dist:
+        base = (
+            args.seed
+            if args.seed is not None
+            else random.SystemRandom().randint(0, 1 << 30)
+        )
+        run_dist(args.iters, base)
+        return
     if args.seed is not None:
         one_case(args.seed)
         print(f"seed {args.seed}: OK")
@@ -194,6 +203,108 @@ def main():
         if (i + 1) % 10 == 0:
             print(f"{i + 1}/{args.iters} ok (base seed {base})")
     print(f"all {args.iters} cases ok (base seed {base})")
+
+
+
+
+# ---------------------------------------------------------------------------
+# distributed fuzzing (world 2, gloo): replicated globs + per-rank leaves
+# stress the partitioner / dedup / manifest-merge machinery
+# ---------------------------------------------------------------------------
+
+
+def _dist_case(seed: int) -> None:
+    import torch.distributed as dist
+
+    from torchsnapshot_amd import Snapshot
+
+    rank = dist.get_rank()
+    rng = random.Random(seed)  # same stream on every rank
+    torch.manual_seed(seed)
+    env = {}
+    if rng.random() < 0.3:
+        env["TSAMD_DISABLE_BATCHING"] = "1"
+    if rng.random() < 0.4:
+        env["TSAMD_MAX_CHUNK_SIZE_BYTES"] = str(rng.choice([256, 8192]))
+    if rng.random() < 0.4:
+        env["TSAMD_SLAB_SIZE_THRESHOLD_BYTES"] = str(rng.choice([256, 1 << 16]))
+    if rng.random() < 0.4:
+        env["TSAMD_CHECKSUM"] = "1"
+        env["TSAMD_VERIFY_CHECKSUM"] = "1"
+    if rng.random() < 0.5:
+        env["TSAMD_ASYNC_SHADOW"] = rng.choice(["0", "1"])
+    old = {k: os.environ.get(k) for k in env}
+    os.environ.update(env)
+    try:
+        # replicated leaves: identical on all ranks (seeded); per-rank
+        # leaves keyed off the rank. Tied aliases included.
+        pool = []
+        sd = {}
+        repl_paths = []
+        for i in range(rng.randint(1, 6)):
+            t = rand_tensor(rng)
+            if pool and rng.random() < 0.2:
+                t = rng.choice(pool)
+            pool.append(t)
+            sd[f"rep{i}"] = t
+            repl_paths.append(f"app/rep{i}")
+        n_local = rng.randint(0, 3)
+        for i in range(n_local):
+            # rank-dependent content AND size
+            torch.manual_seed(seed * 1000 + rank * 7 + i)
+            sd[f"mine{i}"] = torch.rand(rng.randint(1, 50) + rank * 3)
+        expected = {
+            k: (v.clone() if isinstance(v, torch.Tensor) else v)
+            for k, v in sd.items()
+        }
+        holder = Holder(sd)
+        # every rank must agree on the replicated list
+        globs = rng.choice(
+            [repl_paths, ["app/rep*"], ["**/rep*"], repl_paths[:1]]
+        )
+        with tempfile.TemporaryDirectory() as local_d:
+            box = [local_d if rank == 0 else None]
+            dist.broadcast_object_list(box, src=0)
+            path = os.path.join(box[0], "snap")
+            if rng.random() < 0.5:
+                snap = Snapshot.take(path, {"app": holder}, replicated=globs)
+            else:
+                snap = Snapshot.async_take(
+                    path, {"app": holder}, replicated=globs
+                ).wait()
+            out = Holder(
+                {
+                    k: (torch.zeros_like(v) if isinstance(v, torch.Tensor) else None)
+                    for k, v in expected.items()
+                }
+            )
+            snap.restore({"app": out})
+            for k, v in expected.items():
+                eq(v, out.sd[k], f"rank{rank}/{k}")
+            dist.barrier()
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
+
+
+def _dist_worker(seeds) -> None:
+    for s in seeds:
+        try:
+            _dist_case(s)
+        except Exception:
+            print(f"DIST FUZZ FAILURE at seed {s}", file=sys.stderr)
+            raise
+
+
+def run_dist(iters: int, base: int) -> None:
+    from torchsnapshot_amd.test_utils import run_multiprocess
+
+    seeds = [base + i for i in range(iters)]
+    run_multiprocess(2, _dist_worker, seeds)
+    print(f"dist: all {iters} cases ok (base seed {base})")
 
 
 if __name__ == "__main__":

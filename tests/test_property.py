@@ -225,3 +225,29 @@ def test_fuzz_snapshot_seeded(seed):
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     mod.one_case(seed)
+
+
+def _dist_fuzz_seeds(seeds):
+    import importlib.util
+    import os as _os
+
+    spec = importlib.util.spec_from_file_location(
+        "fuzz_snapshot",
+        _os.path.join(
+            _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))),
+            "scripts",
+            "fuzz_snapshot.py",
+        ),
+    )
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod._dist_worker(seeds)
+
+
+def test_fuzz_distributed_seeded():
+    """World-2 slices of the randomized fuzzer: replicated globs,
+    tied aliases, per-rank uneven leaves through partitioner + dedup +
+    manifest merge."""
+    from torchsnapshot_amd.test_utils import run_multiprocess
+
+    run_multiprocess(2, _dist_fuzz_seeds, [77, 7878, 787878])
