@@ -25,12 +25,17 @@ DTYPES = [
 ]
 
 
+USE_GPU = False  # set by --gpu: tensors randomly live on cuda:0
+
+
 def rand_tensor(rng: random.Random):
     dtype = rng.choice(DTYPES)
     ndim = rng.randint(0, 3)
     shape = tuple(rng.randint(0, 9) for _ in range(ndim))
     base = torch.randn(*(s + 2 for s in shape)) if ndim else torch.randn(())
     t = base.to(dtype) if dtype != torch.bool else base > 0
+    if USE_GPU and rng.random() < 0.6:
+        t = t.cuda()
     kind = rng.random()
     if ndim >= 1 and kind < 0.2:
         t = t[tuple(slice(0, s) for s in shape)]  # view into larger
@@ -90,8 +95,8 @@ def rand_state(rng: random.Random, depth=0):
 def eq(a, b, path=""):
     if isinstance(a, torch.Tensor):
         assert isinstance(b, torch.Tensor), f"{path}: {type(b)}"
-        ac = a.detach().contiguous()
-        bc = b.detach().contiguous()
+        ac = a.detach().contiguous().cpu()
+        bc = b.detach().contiguous().cpu()
         assert ac.dtype == bc.dtype, f"{path}: dtype {ac.dtype} vs {bc.dtype}"
         assert ac.shape == bc.shape, f"{path}: shape"
         if ac.numel():
@@ -179,7 +184,11 @@ def main():
     ap.add_argument("--iters", type=int, default=50)
     ap.add_argument("--seed", type=int, default=None)
     ap.add_argument("--dist", action="store_true", help="world-2 gloo fuzz")
+    ap.add_argument("--gpu", action="store_true", help="mix cuda tensors in")
     args = ap.parse_args()
+    if args.gpu:
+        global USE_GPU
+        USE_GPU = True
     if args.dist:
         base = (
             args.seed
