@@ -185,6 +185,7 @@ def main():
     ap.add_argument("--seed", type=int, default=None)
     ap.add_argument("--dist", action="store_true", help="world-2 gloo fuzz")
     ap.add_argument("--gpu", action="store_true", help="mix cuda tensors in")
+    ap.add_argument("--world", type=int, default=2, help="dist world size")
     args = ap.parse_args()
     if args.gpu:
         global USE_GPU
@@ -195,7 +196,7 @@ def main():
             if args.seed is not None
             else random.SystemRandom().randint(0, 1 << 30)
         )
-        run_dist(args.iters, base)
+        run_dist(args.iters, base, args.world)
         return
     if args.seed is not None:
         one_case(args.seed)
@@ -308,12 +309,12 @@ def _dist_worker(seeds) -> None:
             raise
 
 
-def run_dist(iters: int, base: int) -> None:
+def run_dist(iters: int, base: int, world: int = 2) -> None:
     from torchsnapshot_amd.test_utils import run_multiprocess
 
     seeds = [base + i for i in range(iters)]
-    run_multiprocess(2, _dist_worker, seeds)
-    print(f"dist: all {iters} cases ok (base seed {base})")
+    run_multiprocess(world, _dist_worker, seeds)
+    print(f"dist world {world}: all {iters} cases ok (base seed {base})")
 
 
 if __name__ == "__main__":
