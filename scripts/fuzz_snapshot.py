@@ -186,10 +186,26 @@ def main():
     ap.add_argument("--dist", action="store_true", help="world-2 gloo fuzz")
     ap.add_argument("--gpu", action="store_true", help="mix cuda tensors in")
     ap.add_argument("--world", type=int, default=2, help="dist world size")
+    ap.add_argument(
+        "--cross-world",
+        type=str,
+        default=None,
+        metavar="A:B",
+        help="save at world A, restore at world B",
+    )
     args = ap.parse_args()
     if args.gpu:
         global USE_GPU
         USE_GPU = True
+    if args.cross_world:
+        a, b = (int(x) for x in args.cross_world.split(":"))
+        base = (
+            args.seed
+            if args.seed is not None
+            else random.SystemRandom().randint(0, 1 << 30)
+        )
+        run_cross_world(args.iters, base, a, b)
+        return
     if args.dist:
         base = (
             args.seed
@@ -315,6 +331,72 @@ def run_dist(iters: int, base: int, world: int = 2) -> None:
     seeds = [base + i for i in range(iters)]
     run_multiprocess(world, _dist_worker, seeds)
     print(f"dist world {world}: all {iters} cases ok (base seed {base})")
+
+
+
+
+# ---------------------------------------------------------------------------
+# cross-world elasticity fuzzing: save replicated state at world A,
+# restore at world B (borrowing for B > A, subset reads for B < A)
+# ---------------------------------------------------------------------------
+
+
+def _xw_state(seed: int):
+    rng = random.Random(seed)
+    torch.manual_seed(seed)
+    pool = []
+    sd = {}
+    for i in range(rng.randint(1, 6)):
+        t = rand_tensor(rng)
+        if pool and rng.random() < 0.2:
+            t = rng.choice(pool)
+        pool.append(t)
+        sd[f"rep{i}"] = t
+    return sd, rng
+
+
+def _xw_save(seeds, shared_dir) -> None:
+    from torchsnapshot_amd import Snapshot
+
+    for s in seeds:
+        sd, rng = _xw_state(s)
+        globs = rng.choice([["**"], [f"app/rep{i}" for i in range(len(sd))]])
+        path = os.path.join(shared_dir, f"snap_{s}")
+        if rng.random() < 0.5:
+            Snapshot.take(path, {"app": Holder(sd)}, replicated=globs)
+        else:
+            Snapshot.async_take(
+                path, {"app": Holder(sd)}, replicated=globs
+            ).wait()
+
+
+def _xw_restore(seeds, shared_dir) -> None:
+    from torchsnapshot_amd import Snapshot
+
+    for s in seeds:
+        expected, _ = _xw_state(s)
+        out = Holder(
+            {
+                k: (torch.zeros_like(v) if isinstance(v, torch.Tensor) else None)
+                for k, v in expected.items()
+            }
+        )
+        Snapshot(os.path.join(shared_dir, f"snap_{s}")).restore({"app": out})
+        for k, v in expected.items():
+            eq(v, out.sd[k], f"xw/{k}")
+
+
+def run_cross_world(iters: int, base: int, w_save: int, w_restore: int) -> None:
+    from torchsnapshot_amd.test_utils import run_multiprocess
+
+    seeds = [base + i for i in range(iters)]
+    with tempfile.TemporaryDirectory() as d:
+        run_multiprocess(w_save, _xw_save, seeds, d)
+        run_multiprocess(w_restore, _xw_restore, seeds, d)
+    print(
+        f"cross-world {w_save}->{w_restore}: all {iters} cases ok "
+        f"(base seed {base})"
+    )
 
 
 if __name__ == "__main__":

@@ -251,3 +251,23 @@ def test_fuzz_distributed_seeded():
     from torchsnapshot_amd.test_utils import run_multiprocess
 
     run_multiprocess(2, _dist_fuzz_seeds, [77, 7878, 787878])
+
+
+def test_fuzz_cross_world_seeded():
+    """Elasticity fuzz slice: replicated state saved at world 2 restores
+    bit-exactly at world 3 (borrowing) and vice versa."""
+    import importlib.util
+    import os as _os
+
+    spec = importlib.util.spec_from_file_location(
+        "fuzz_snapshot",
+        _os.path.join(
+            _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))),
+            "scripts",
+            "fuzz_snapshot.py",
+        ),
+    )
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.run_cross_world(2, 4242, 2, 3)
+    mod.run_cross_world(2, 2424, 3, 2)
