@@ -253,9 +253,7 @@ def test_fuzz_distributed_seeded():
     run_multiprocess(2, _dist_fuzz_seeds, [77, 7878, 787878])
 
 
-def test_fuzz_cross_world_seeded():
-    """Elasticity fuzz slice: replicated state saved at world 2 restores
-    bit-exactly at world 3 (borrowing) and vice versa."""
+def _xw_phase(phase, seeds, d):
     import importlib.util
     import os as _os
 
@@ -269,5 +267,18 @@ def test_fuzz_cross_world_seeded():
     )
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
-    mod.run_cross_world(2, 4242, 2, 3)
-    mod.run_cross_world(2, 2424, 3, 2)
+    (mod._xw_save if phase == "save" else mod._xw_restore)(seeds, d)
+
+
+def test_fuzz_cross_world_seeded():
+    """Elasticity fuzz slice: replicated state saved at world 2 restores
+    bit-exactly at world 3 (borrowing) and vice versa."""
+    import tempfile as _tf
+
+    from torchsnapshot_amd.test_utils import run_multiprocess
+
+    for w_save, w_restore, base in ((2, 3, 4242), (3, 2, 2424)):
+        seeds = [base, base + 1]
+        with _tf.TemporaryDirectory() as d:
+            run_multiprocess(w_save, _xw_phase, "save", seeds, d)
+            run_multiprocess(w_restore, _xw_phase, "restore", seeds, d)
