@@ -58,33 +58,67 @@ def psum64_hexdigest(buf, word_base: int = 0) -> str:
     """psum64 of ``buf``. ``word_base`` is the u64 word index of the
     buffer's first byte within the containing file, so subrange checksums
     of byte-range reads line up with the file-global indexing used by the
-    device kernel (ops/hip/staging.hip)."""
-    import numpy as np
+    device kernel (ops/hip/staging.hip).
+
+    Fast path: a single-pass threaded C++ loop in the native extension
+    (memcpy-rate; the vectorized python implementations burn ~10 memory
+    passes on splitmix temporaries and manage only ~0.2-0.3 GB/s).
+    Falls back to torch int64 ops (two's-complement wraparound IS mod
+    2^64 arithmetic) when the extension is unavailable."""
+    import torch
 
     mv = memoryview(buf)
     if mv.format != "B":
         mv = mv.cast("B")
-    total = np.uint64(0)
-    chunk_words = 4 * 1024 * 1024  # 32 MB pieces keep index arrays small
     nbytes = mv.nbytes
-    word_base = np.uint64(word_base)
-    off = 0
-    old = np.seterr(over="ignore")
+    if nbytes == 0:
+        return "psum64:" + format(0, "016x")
     try:
-        while off < nbytes:
-            end = min(off + chunk_words * 8, nbytes)
-            piece = np.frombuffer(mv[off:end], dtype=np.uint8)
-            pad = (-len(piece)) % 8
-            if pad:
-                piece = np.concatenate([piece, np.zeros(pad, dtype=np.uint8)])
-            words = piece.view("<u8")
-            idx = np.arange(word_base, word_base + len(words), dtype=np.uint64)
-            total = total + np.uint64((words * _splitmix64_mult(idx)).sum())
-            word_base += len(words)
-            off = end
-    finally:
-        np.seterr(**old)
-    return "psum64:" + format(int(total), "016x")
+        from . import _csnap
+
+        return "psum64:" + format(
+            _csnap.psum64_host(mv, word_base) % (1 << 64), "016x"
+        )
+    except ImportError:
+        pass
+    total = 0
+    chunk_words = 16 * 1024 * 1024  # 128 MB pieces
+    off = 0
+    wb = word_base
+    while off < nbytes:
+        end = min(off + chunk_words * 8, nbytes)
+        piece = torch.frombuffer(mv[off:end], dtype=torch.uint8)
+        pad = (-piece.numel()) % 8
+        if pad:
+            piece = torch.cat([piece, torch.zeros(pad, dtype=torch.uint8)])
+        words = piece.view(torch.int64)
+        idx = torch.arange(wb, wb + words.numel(), dtype=torch.int64)
+        total = (total + _psum_torch_chunk(words, idx)) % (1 << 64)
+        wb += words.numel()
+        off = end
+    return "psum64:" + format(total, "016x")
+
+
+def _psum_torch_chunk(words: "torch.Tensor", idx: "torch.Tensor") -> int:
+    """sum(words * (splitmix64(idx) | 1)) mod 2^64 with int64 wraparound.
+    Right shifts must be LOGICAL (the values are bit patterns, not signed
+    numbers): torch >> on int64 is arithmetic, so mask the copied sign
+    bits off after shifting."""
+    import torch
+
+    def lshr(v: torch.Tensor, n: int) -> torch.Tensor:
+        return (v >> n) & ((1 << (64 - n)) - 1)
+
+    def c(val: int) -> int:  # u64 constant -> i64 two's complement
+        return val - (1 << 64) if val >= (1 << 63) else val
+
+    z = idx + c(0x9E3779B97F4A7C15)
+    z = (z ^ lshr(z, 30)) * c(0xBF58476D1CE4E5B9)
+    z = (z ^ lshr(z, 27)) * c(0x94D049BB133111EB)
+    z = z ^ lshr(z, 31)
+    mult = z | 1
+    s = int((words * mult).sum().item())
+    return s % (1 << 64)
 
 
 def checksum_file_path(rank: int) -> str:
