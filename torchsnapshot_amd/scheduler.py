@@ -261,8 +261,14 @@ def execute_write_reqs(
                     if pre is not None:
                         checksums[req.path] = pre
                     else:
+                        # psum64 for host-staged payloads too (native
+                        # single-pass hash, ~14 GB/s): every file shares
+                        # one algorithm, so byte-range/tiled reads of ANY
+                        # payload are verifiable on restore
                         checksums[req.path] = await asyncio.get_running_loop(
-                        ).run_in_executor(executor, integrity.hash_buffer, buf)
+                        ).run_in_executor(
+                            executor, integrity.psum64_hexdigest, buf
+                        )
                     # per-member values so byte-range/merged-span reads of
                     # batched slabs are verifiable on restore
                     for s, e, v in (
