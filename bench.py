@@ -12,8 +12,10 @@ Launch (driver contract):
     python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
         --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
 
-At N>1 each rank holds 1/N of every parameter as a DTensor on a 1-D mesh
-(FSDP2-style layout); at N=1 parameters are plain device tensors.
+At every N (including 1) each rank holds 1/N of every parameter as a
+DTensor on a 1-D mesh (FSDP2-style layout), so the FSDP write path named
+by the metric is always the one timed; if the 1-rank process group can't
+come up, N=1 falls back to plain device tensors (parallelism "single").
 """
 
 from __future__ import annotations
@@ -23,7 +25,7 @@ import json
 import os
 import shutil
 import time
-from typing import Dict, List, Tuple
+from typing import Dict, List, Optional, Tuple
 
 import torch
 
@@ -87,11 +89,14 @@ def build_state(
     world_size: int,
     dtype: torch.dtype,
     model: str = "llama3-8b",
+    use_dtensor: Optional[bool] = None,
 ) -> Tuple[_BenchState, int]:
+    if use_dtensor is None:
+        use_dtensor = world_size > 1
     shapes = tiny_shapes() if model == "tiny" else llama3_8b_shapes()
     total_bytes = 0
     sd: Dict[str, torch.Tensor] = {}
-    if world_size > 1:
+    if use_dtensor:
         from torch.distributed.device_mesh import init_device_mesh
         from torch.distributed.tensor import DTensor
         from torch.distributed.tensor.placement_types import Shard
@@ -181,6 +186,29 @@ def main() -> None:
     else:
         device = torch.device("cpu")
 
+    # at world 1, still run the FSDP/DTensor write path (the metric's
+    # named parallelism) on a 1-rank mesh; fall back to plain tensors if
+    # the single-rank process group cannot come up
+    use_dtensor = world_size > 1
+    single_rank_pg = False
+    if world_size == 1 and args.model != "tiny":
+        import torch.distributed as dist
+
+        try:
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29561")
+            kwargs = {"device_id": device} if use_cuda else {}
+            dist.init_process_group(
+                backend="nccl" if use_cuda else "gloo",
+                rank=0,
+                world_size=1,
+                **kwargs,
+            )
+            use_dtensor = True
+            single_rank_pg = True
+        except Exception:
+            use_dtensor = False
+
     bench_dir = args.dir or "/tmp/tsamd_bench"
     ckpt_path = os.path.join(bench_dir, "ckpt")
     if rank == 0:
@@ -189,7 +217,11 @@ def main() -> None:
     _barrier(world_size)
 
     state, total_bytes = build_state(
-        device, world_size, torch.bfloat16, model=args.model
+        device,
+        world_size,
+        torch.bfloat16,
+        model=args.model,
+        use_dtensor=use_dtensor,
     )
     app_state = {"model": state}
 
@@ -258,13 +290,13 @@ def main() -> None:
                 "model_bytes": total_bytes,
                 "global_batch": None,
                 "seq_len": None,
-                "parallelism": f"fsdp{world_size}" if world_size > 1 else "single",
+                "parallelism": f"fsdp{world_size}" if use_dtensor else "single",
                 "storage": bench_dir,
             },
         }
         print(json.dumps(result))
 
-    if world_size > 1:
+    if world_size > 1 or single_rank_pg:
         import torch.distributed as dist
 
         dist.destroy_process_group()
