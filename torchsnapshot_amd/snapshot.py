@@ -495,11 +495,16 @@ class Snapshot:
                         idx, free / 1e9, need / 1e9,
                     )
                     return False
-        # clone, preserving aliasing (tied weights -> one shared clone)
+        # clone, preserving aliasing (tied weights -> one shared clone).
+        # DTensors are rebuilt with from_local around a plain local clone:
+        # DTensor.clone() dispatches through the tensor-parallel op layer
+        # per call and cost ~0.6 ms x hundreds of params in the stall
+        # window; from_local with explicit shape/stride is metadata-only.
         clones: Dict[Any, Any] = {}
         for p in tensor_paths:
             obj = flattened[p]
-            local = obj.to_local() if (DTensor and isinstance(obj, DTensor)) else obj
+            is_dt = DTensor and isinstance(obj, DTensor)
+            local = obj.to_local() if is_dt else obj
             key = (
                 local.data_ptr(),
                 local.dtype,
@@ -508,7 +513,17 @@ class Snapshot:
                 str(local.device),
             )
             if key not in clones:
-                clones[key] = obj.detach().clone()
+                if is_dt:
+                    clones[key] = DTensor.from_local(
+                        local.detach().clone(),
+                        obj.device_mesh,
+                        obj.placements,
+                        run_check=False,
+                        shape=obj.shape,
+                        stride=obj.stride(),
+                    )
+                else:
+                    clones[key] = obj.detach().clone()
             flattened[p] = clones[key]
         for idx in cuda_need:
             torch.cuda.synchronize(idx)
