@@ -57,10 +57,34 @@ def _local_global_offset(dt: Any) -> Tuple[List[int], List[int]]:
     return list(local_shape), list(global_offset)
 
 
+_replica_set_cache: dict = {}
+
+
 def _my_replica_set(dt: Any) -> List[int]:
     """Global ranks that hold a shard identical to this rank's (the ranks
     reached by varying only replicated mesh dims at this rank's mesh
-    coordinate)."""
+    coordinate). Cached per (mesh, placements): a model checkpoint calls
+    this once per parameter with identical inputs, and the mesh tensor
+    math is measurable in the async-stall window under CPU contention."""
+    from torch.distributed.tensor.placement_types import Shard
+
+    mesh = dt.device_mesh.mesh
+    cache_key = (
+        tuple(mesh.flatten().tolist()),
+        tuple(mesh.shape),
+        tuple(str(p) for p in dt.placements),
+        torch.distributed.get_rank() if torch.distributed.is_initialized() else 0,
+    )
+    hit = _replica_set_cache.get(cache_key)
+    if hit is not None:
+        return hit
+    result = _my_replica_set_impl(dt)
+    if len(_replica_set_cache) < 64:
+        _replica_set_cache[cache_key] = result
+    return result
+
+
+def _my_replica_set_impl(dt: Any) -> List[int]:
     from torch.distributed.tensor.placement_types import Shard
 
     mesh = dt.device_mesh.mesh
