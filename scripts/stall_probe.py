@@ -1,4 +1,6 @@
 #!/usr/bin/env python3
+# NOTE: experimental probe; the monkeypatch pattern is validated on CPU
+# but one slow GPU box timed out running it — not part of any suite.
 """Decompose the async_take stall on a GPU box (DTensor world-1 path)."""
 import os
 import sys
