@@ -147,3 +147,44 @@ def test_run_coro_sync_nested():
 
     assert asyncio.run(outer()) == 42
     assert run_coro_sync(inner()) == 42
+
+
+def test_concurrent_snapshots_from_threads():
+    """Two threads taking/restoring snapshots to different paths at once:
+    the per-call pipeline threads, shared pools, and storage executors
+    must not interfere."""
+    import tempfile
+    import threading
+
+    import torch
+
+    from torchsnapshot_amd import Snapshot, StateDict
+
+    errors = []
+
+    def worker(idx: int, d: str) -> None:
+        try:
+            torch.manual_seed(idx)
+            sd = StateDict(
+                **{f"t{i}": torch.rand(64, 32) for i in range(6)}, n=idx
+            )
+            path = f"{d}/snap{idx}"
+            snap = Snapshot.take(path, {"sd": sd})
+            out = StateDict()
+            snap.restore({"sd": out})
+            for k in sd:
+                if isinstance(sd[k], torch.Tensor):
+                    assert torch.equal(out[k], sd[k]), (idx, k)
+            assert out["n"] == idx
+        except Exception as e:  # noqa: BLE001
+            errors.append((idx, e))
+
+    with tempfile.TemporaryDirectory() as d:
+        threads = [
+            threading.Thread(target=worker, args=(i, d)) for i in range(4)
+        ]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+    assert not errors, errors
