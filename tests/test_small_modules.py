@@ -202,3 +202,76 @@ def test_rand_tensor_dtypes():
         assert t.dtype == dtype and t.shape == (4, 4)
     q = rand_tensor((8,), torch.qint8)
     assert q.is_quantized
+
+
+# ---------------------------------------------------------------------------
+# third-party storage plugin discovery (entry points)
+# ---------------------------------------------------------------------------
+
+
+class _RamPlugin:
+    """Minimal in-memory StoragePlugin registered via a fake entry point."""
+
+    store = {}
+
+    def __init__(self, root, storage_options=None):
+        self.root = root
+
+    async def write(self, write_io):
+        self.store[f"{self.root}/{write_io.path}"] = bytes(
+            memoryview(write_io.buf)
+        )
+
+    async def read(self, read_io):
+        data = self.store[f"{self.root}/{read_io.path}"]
+        if read_io.byte_range is not None:
+            s, e = read_io.byte_range
+            data = data[s:e]
+        read_io.buf = bytearray(data)
+
+    async def delete(self, path):
+        self.store.pop(f"{self.root}/{path}", None)
+
+    async def delete_dir(self, path):
+        prefix = f"{self.root}/{path}"
+        for k in [k for k in self.store if k.startswith(prefix)]:
+            del self.store[k]
+
+
+def test_third_party_storage_plugin_entry_point(monkeypatch):
+    """A `tsamd_storage_plugins` entry point resolves custom protocols
+    (parity with reference storage_plugin.py:55-67) and a full snapshot
+    round-trips through it."""
+    import torch
+
+    import torchsnapshot_amd.storage as storage_mod
+    from torchsnapshot_amd import Snapshot, StateDict
+    from torchsnapshot_amd.io_types import StoragePlugin
+
+    class _EP:
+        name = "ram"
+
+        @staticmethod
+        def load():
+            return _RamPlugin
+
+    def fake_entry_points(group=None):
+        assert group == "tsamd_storage_plugins"
+        return [_EP]
+
+    monkeypatch.setattr(storage_mod, "entry_points", fake_entry_points)
+    # make sure the ABC sync wrappers exist on the minimal plugin
+    for name in ("sync_write", "sync_read", "sync_close", "close",
+                 "close_for_loop"):
+        assert hasattr(StoragePlugin, name)
+    _RamPlugin.sync_write = StoragePlugin.sync_write
+    _RamPlugin.sync_read = StoragePlugin.sync_read
+    _RamPlugin.sync_close = StoragePlugin.sync_close
+    _RamPlugin.close = StoragePlugin.close
+    _RamPlugin.close_for_loop = StoragePlugin.close_for_loop
+
+    sd = StateDict(w=torch.rand(32, 8), n=5)
+    snap = Snapshot.take("ram://bucket/ckpt", {"sd": sd})
+    out = StateDict()
+    snap.restore({"sd": out})
+    assert torch.equal(out["w"], sd["w"]) and out["n"] == 5
